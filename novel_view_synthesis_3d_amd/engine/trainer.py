@@ -1,0 +1,215 @@
+"""Trainer — the reference train.py Trainer re-designed for MI355X.
+
+Keeps the reference's public surface (Trainer(folder, train_batch_size=...,
+...).train(), reference train.py:78-127) and fixes its defects:
+  * true DP instead of the pmap ensemble (D3) — parallel/ddp.py
+  * per-step CFG mask + dropout randomness (D2: the reference bakes one mask
+    at trace time)
+  * on-device forward noising (K19 moved out of dataset workers)
+  * resumable checkpoints with optimizer state (§5.4)
+  * per-element MSE loss by default; `loss='frob'` reproduces the
+    reference's whole-batch Frobenius-norm quirk (D8, train.py:67)
+  * bf16 autocast on GPU; fp32 master weights + Adam
+"""
+
+from __future__ import annotations
+
+import contextlib
+import math
+import os
+import time
+from pathlib import Path
+from typing import Dict, Optional
+
+import torch
+import torch.nn.functional as F
+
+from novel_view_synthesis_3d_amd.config import TrainConfig, XUNetConfig
+from novel_view_synthesis_3d_amd.data.srn import SceneClassDataset
+from novel_view_synthesis_3d_amd.data.synthetic import synthetic_batch
+from novel_view_synthesis_3d_amd.diffusion.forward import q_sample
+from novel_view_synthesis_3d_amd.diffusion.schedules import DiffusionSchedule
+from novel_view_synthesis_3d_amd.engine import checkpoint as ckpt
+from novel_view_synthesis_3d_amd.engine.metrics import JsonlLogger, RateMeter
+from novel_view_synthesis_3d_amd.models.xunet import XUNet
+from novel_view_synthesis_3d_amd.parallel.ddp import (
+    DataParallelEngine, init_distributed,
+)
+
+
+def _cycle(dl):
+    while True:
+        for d in dl:
+            yield d
+
+
+class Trainer:
+    def __init__(self, folder: Optional[str] = None, *,
+                 train_batch_size: int = 2,
+                 train_lr: float = 1e-4,
+                 train_num_steps: int = 100_000,
+                 save_every: int = 1000,
+                 img_sidelength: int = 64,
+                 results_folder: str = "./results",
+                 model_cfg: Optional[XUNetConfig] = None,
+                 train_cfg: Optional[TrainConfig] = None,
+                 device: Optional[str] = None):
+        cfg = train_cfg or TrainConfig()
+        cfg.train_batch_size = train_batch_size
+        cfg.train_lr = train_lr
+        cfg.train_num_steps = train_num_steps
+        cfg.save_every = save_every
+        cfg.img_sidelength = img_sidelength
+        cfg.results_folder = results_folder
+        self.cfg = cfg
+        self.model_cfg = model_cfg or XUNetConfig()
+
+        # --- distributed / device -------------------------------------
+        self.rank, self.world, self.local_rank = init_distributed()
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        if device == "cuda":
+            torch.cuda.set_device(self.local_rank)
+            device = f"cuda:{self.local_rank}"
+        self.device = torch.device(device)
+        self.is_main = self.rank == 0
+
+        torch.manual_seed(cfg.seed * 9176 + self.rank)
+
+        # --- model / optimizer / DP -----------------------------------
+        self.model = XUNet(self.model_cfg, img_sidelength).to(self.device)
+        self.model.train()
+        self.opt = torch.optim.Adam(self.model.parameters(), lr=cfg.train_lr,
+                                    betas=cfg.adam_betas, eps=cfg.adam_eps)
+        self.ddp = DataParallelEngine(self.model, bucket_mb=cfg.bucket_mb)
+        self.schedule = DiffusionSchedule(1000)
+        self.step = 0
+
+        # --- data ------------------------------------------------------
+        self.folder = folder
+        mode = cfg.data
+        if mode == "auto":
+            mode = "srn" if (folder and os.path.isdir(folder)) else "synthetic"
+        self.data_mode = mode
+        self.dl = None
+        if mode == "srn":
+            ds = SceneClassDataset(root_dir=folder,
+                                   max_num_instances=-1,
+                                   max_observations_per_instance=50,
+                                   img_sidelength=img_sidelength,
+                                   specific_observation_idcs=None,
+                                   samples_per_instance=1)
+            assert len(ds) > 0
+            sampler = None
+            if self.world > 1:
+                sampler = torch.utils.data.distributed.DistributedSampler(ds)
+            self.dl = _cycle(torch.utils.data.DataLoader(
+                ds, batch_size=train_batch_size, shuffle=(sampler is None),
+                sampler=sampler, drop_last=True, collate_fn=ds.collate_fn,
+                pin_memory=(self.device.type == "cuda"),
+                num_workers=cfg.num_workers))
+        self._gen = torch.Generator(device=self.device)
+        self._gen.manual_seed(cfg.seed * 131071 + self.rank + 1)
+
+        # --- bookkeeping ----------------------------------------------
+        self.results_folder = Path(results_folder)
+        if self.is_main:
+            self.results_folder.mkdir(exist_ok=True, parents=True)
+        self.logger = JsonlLogger(
+            str(self.results_folder / "train_log.jsonl"), enabled=self.is_main)
+        self.meter = RateMeter()
+
+        if cfg.resume:
+            path = (ckpt.find_latest(cfg.ckpt_folder)
+                    if cfg.resume == "auto" else cfg.resume)
+            if path:
+                self.step = ckpt.load_checkpoint(path, self.model, self.opt,
+                                                 map_location=self.device)
+                if self.is_main:
+                    print(f"resumed from {path} at step {self.step}")
+
+    # -----------------------------------------------------------------
+    def next_batch(self) -> Dict[str, torch.Tensor]:
+        """A raw batch (x, x_target, poses) on device."""
+        if self.data_mode == "synthetic":
+            return synthetic_batch(self.cfg.train_batch_size,
+                                   self.cfg.img_sidelength,
+                                   device=self.device, generator=self._gen)
+        obs, _ = next(self.dl)
+        return {k: v.to(self.device, non_blocking=True) for k, v in obs.items()}
+
+    def prepare_model_inputs(self, raw: Dict[str, torch.Tensor]):
+        """On-device forward noising (K19) + CFG mask (per-step randomness,
+        fixing reference D2)."""
+        B = raw["x"].shape[0]
+        if "z" in raw:  # cpu_noising dataset path
+            batch = dict(raw)
+            noise = raw["noise"]
+        else:
+            t = torch.randint(0, self.schedule.timesteps, (B,),
+                              device=self.device, generator=self._gen)
+            z, noise, logsnr = q_sample(raw["x_target"], t, self.schedule,
+                                        generator=self._gen)
+            batch = {"x": raw["x"], "z": z, "logsnr": logsnr.to(self.device),
+                     "R1": raw["R1"], "t1": raw["t1"],
+                     "R2": raw["R2"], "t2": raw["t2"], "K": raw["K"]}
+        cond_mask = (torch.rand(B, device=self.device, generator=self._gen)
+                     > self.cfg.cond_drop_prob).to(torch.float32)
+        return batch, cond_mask, noise
+
+    def compute_loss(self, out: torch.Tensor, noise: torch.Tensor):
+        out = out.to(torch.float32)
+        noise = noise.to(torch.float32)
+        if self.cfg.loss == "frob":  # reference train.py:67 (quirk D8)
+            return torch.linalg.vector_norm(out - noise)
+        return F.mse_loss(out, noise)
+
+    def _autocast(self):
+        if self.device.type == "cuda" and self.cfg.amp == "bf16":
+            return torch.autocast("cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
+    def train_step(self, raw: Optional[Dict[str, torch.Tensor]] = None
+                   ) -> torch.Tensor:
+        """One optimization step; returns the (local) loss tensor."""
+        if raw is None:
+            raw = self.next_batch()
+        batch, cond_mask, noise = self.prepare_model_inputs(raw)
+        self.ddp.zero_flags()
+        self.opt.zero_grad(set_to_none=True)
+        with self._autocast():
+            out = self.model(batch, cond_mask)
+        loss = self.compute_loss(out, noise)
+        loss.backward()
+        self.ddp.finish()
+        self.opt.step()
+        return loss.detach()
+
+    def train(self) -> None:
+        cfg = self.cfg
+        while self.step < cfg.train_num_steps:
+            t0 = time.perf_counter()
+            loss = self.train_step()
+            self.meter.update(cfg.train_batch_size * self.world)
+            if self.step % cfg.log_every == 0:
+                lval = float(self.ddp.all_reduce_scalar(loss).item()) \
+                    if self.ddp.enabled else float(loss.item())
+                if self.is_main:
+                    rate = self.meter.rate()
+                    print(f"{self.step}: loss={lval:.5f} "
+                          f"images/sec={rate:.1f}", flush=True)
+                    self.logger.log(step=self.step, loss=lval,
+                                    images_per_sec=rate,
+                                    step_time=time.perf_counter() - t0)
+            if self.step % cfg.save_every == 0 and self.step > 0 and self.is_main:
+                ckpt.save_checkpoint(
+                    cfg.ckpt_folder, self.model, self.opt, self.step,
+                    extra={"model_cfg": vars(self.model_cfg),
+                           "img_sidelength": cfg.img_sidelength})
+            self.step += 1
+        if self.is_main:
+            ckpt.save_checkpoint(
+                cfg.ckpt_folder, self.model, self.opt, self.step,
+                extra={"model_cfg": vars(self.model_cfg),
+                       "img_sidelength": cfg.img_sidelength})
+            print("training completed")
