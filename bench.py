@@ -97,6 +97,9 @@ def main() -> None:
                 "parallelism": f"dp{n_gpus}",
                 "final_loss": float(loss.item()),
                 "params": trainer.model.num_params(),
+                "peak_mem_gb": round(
+                    torch.cuda.max_memory_allocated() / 1e9, 2)
+                if use_cuda else None,
             },
         }
         print(json.dumps(result), flush=True)
