@@ -79,19 +79,15 @@ class ConditioningProcessor(nn.Module):
         logsnr_emb = logsnr_emb.to(batch["x"].dtype)
         logsnr_emb = self.Dense_1(F.silu(self.Dense_0(logsnr_emb)))
 
-        # pose embeddings (K13 + K14)
-        pose_emb = []
-        for Rk, tk in (("R1", "t1"), ("R2", "t2")):
-            pos, direc = camera_rays(batch[Rk], batch[tk], batch["K"], H, W)
-            pe = torch.cat([ops.posenc_nerf(pos, 0, 15),
-                            ops.posenc_nerf(direc, 0, 8)], dim=-1)
-            pose_emb.append(pe)
-        pose_emb = torch.stack(pose_emb, dim=1).to(batch["x"].dtype)  # (B,2,H,W,144)
-
-        # CFG masking (K15)
+        # pose embeddings + CFG masking (K13+K14+K15, one fused kernel on GPU)
         assert cond_mask.shape == (B,), cond_mask.shape
-        mask = cond_mask.to(pose_emb.dtype).reshape(B, 1, 1, 1, 1)
-        pose_emb = pose_emb * mask
+        emb_dtype = batch["x"].dtype
+        if batch["x"].is_cuda and torch.is_autocast_enabled():
+            emb_dtype = torch.get_autocast_dtype("cuda")
+        R2 = torch.stack([batch["R1"], batch["R2"]], dim=1)
+        t2 = torch.stack([batch["t1"], batch["t2"]], dim=1)
+        pose_emb = ops.pose_embedding(R2, t2, batch["K"], cond_mask,
+                                      H, W, emb_dtype)  # (B,2,H,W,144)
         if cfg.use_pos_emb:
             pose_emb = pose_emb + self.pos_emb[None, None]
         if cfg.use_ref_pose_emb:

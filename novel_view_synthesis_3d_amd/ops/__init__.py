@@ -108,3 +108,9 @@ def residual_scale_add(h, h_in):
     if _use_hip(h, "residual_scale_add"):
         return _HIP_MOD.residual_scale_add(h, h_in)
     return ref.residual_scale_add(h, h_in)
+
+
+def pose_embedding(R, t, K, cond_mask, H: int, W: int, out_dtype):
+    if _use_hip(R, "pose_embedding"):
+        return _HIP_MOD.pose_embedding(R, t, K, cond_mask, H, W, out_dtype)
+    return ref.pose_embedding(R, t, K, cond_mask, H, W, out_dtype)

@@ -19,8 +19,9 @@ EXT_NAME = "nvs3d_hip"
 
 
 def hip_sources():
-    return sorted(glob.glob(os.path.join(HIP_DIR, "*.hip"))) + \
-        sorted(glob.glob(os.path.join(HIP_DIR, "*.cpp")))
+    hips = [s for s in sorted(glob.glob(os.path.join(HIP_DIR, "*.hip")))
+            if not s.endswith("_hip.hip")]  # hipify-generated copies
+    return hips + sorted(glob.glob(os.path.join(HIP_DIR, "*.cpp")))
 
 
 def built_path() -> str:

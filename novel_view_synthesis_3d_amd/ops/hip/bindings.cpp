@@ -1,0 +1,47 @@
+// torch op registration for the nvs3d gfx950 HIP kernels.
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta,
+                                  c10::optional<torch::Tensor> fscale,
+                                  c10::optional<torch::Tensor> fshift,
+                                  int64_t groups, double eps, bool silu);
+std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor gamma, torch::Tensor beta,
+                                  c10::optional<torch::Tensor> fscale,
+                                  c10::optional<torch::Tensor> fshift,
+                                  torch::Tensor mean, torch::Tensor rstd,
+                                  int64_t groups, bool silu);
+torch::Tensor rays_posenc(torch::Tensor R, torch::Tensor t, torch::Tensor Kinv,
+                          c10::optional<torch::Tensor> mask,
+                          int64_t H, int64_t W, torch::ScalarType out_dtype);
+void fused_adam(torch::Tensor ptrs, torch::Tensor chunk_tensor,
+                torch::Tensor chunk_off, torch::Tensor numels,
+                double lr, double b1, double b2, double eps, int64_t step);
+
+torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
+                             torch::Tensor Kinv,
+                             c10::optional<torch::Tensor> mask,
+                             int64_t H, int64_t W, torch::Tensor dtype_like) {
+  return rays_posenc(R, t, Kinv, mask, H, W, dtype_like.scalar_type());
+}
+
+TORCH_LIBRARY(nvs3d, m) {
+  m.def("gn_fwd(Tensor x, Tensor gamma, Tensor beta, Tensor? fscale, "
+        "Tensor? fshift, int groups, float eps, bool silu) -> Tensor[]");
+  m.def("gn_bwd(Tensor dy, Tensor x, Tensor gamma, Tensor beta, "
+        "Tensor? fscale, Tensor? fshift, Tensor mean, Tensor rstd, "
+        "int groups, bool silu) -> Tensor[]");
+  m.def("rays_posenc(Tensor R, Tensor t, Tensor Kinv, Tensor? mask, "
+        "int H, int W, Tensor dtype_like) -> Tensor");
+  m.def("fused_adam(Tensor ptrs, Tensor chunk_tensor, Tensor chunk_off, "
+        "Tensor numels, float lr, float b1, float b2, float eps, "
+        "int step) -> ()");
+}
+
+TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
+  m.impl("gn_fwd", gn_fwd);
+  m.impl("gn_bwd", gn_bwd);
+  m.impl("rays_posenc", rays_posenc_py);
+  m.impl("fused_adam", fused_adam);
+}

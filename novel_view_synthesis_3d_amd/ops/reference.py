@@ -146,3 +146,23 @@ def posenc_nerf(x: torch.Tensor, min_deg: int = 0, max_deg: int = 15) -> torch.T
                           dtype=x.dtype, device=x.device)
     xb = (x[..., None, :] * scales[:, None]).reshape(*x.shape[:-1], -1)
     return torch.cat([x, torch.sin(xb), torch.sin(xb + math.pi / 2.0)], dim=-1)
+
+
+def pose_embedding(R: torch.Tensor, t: torch.Tensor, K: torch.Tensor,
+                   cond_mask: Optional[torch.Tensor], H: int, W: int,
+                   out_dtype: torch.dtype) -> torch.Tensor:
+    """K13+K14+K15 fused: per-pixel rays for BOTH cameras -> NeRF posenc ->
+    CFG mask. R: (B,2,3,3), t: (B,2,3), K: (B,3,3). Returns (B,2,H,W,144).
+
+    Eager oracle for the rays_posenc HIP kernel
+    (reference model/xunet.py:159-179)."""
+    from novel_view_synthesis_3d_amd.models.rays import camera_rays
+    embs = []
+    for f in range(2):
+        pos, direc = camera_rays(R[:, f], t[:, f], K, H, W)
+        embs.append(torch.cat([posenc_nerf(pos, 0, 15),
+                               posenc_nerf(direc, 0, 8)], dim=-1))
+    pe = torch.stack(embs, dim=1)
+    if cond_mask is not None:
+        pe = pe * cond_mask.to(pe.dtype).reshape(-1, 1, 1, 1, 1)
+    return pe.to(out_dtype)

@@ -1,0 +1,164 @@
+"""HIP kernel parity tests vs the eager fp32 oracle (ops/reference.py).
+
+All tests run on a real MI355X (pytest -m gpu via gpurun). Policy: every
+kernel is compared against the plain-PyTorch fp32 reference of the same op;
+bf16 paths get bf16-scale tolerances.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from novel_view_synthesis_3d_amd.ops import hip_ops
+    from novel_view_synthesis_3d_amd.ops import reference as ref
+
+
+def _gn_case(B, F, H, W, C, dtype, film, silu, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    x = torch.randn(B, F, H, W, C, device="cuda", generator=g, dtype=dtype)
+    gamma = (torch.randn(C, device="cuda", generator=g) * 0.1 + 1.0)
+    beta = torch.randn(C, device="cuda", generator=g) * 0.1
+    fs = ft = None
+    if film:
+        fs = torch.randn(B, F, H, W, C, device="cuda", generator=g,
+                         dtype=dtype) * 0.2
+        ft = torch.randn(B, F, H, W, C, device="cuda", generator=g,
+                         dtype=dtype) * 0.2
+    return x, gamma, beta, fs, ft
+
+
+GN_SHAPES = [
+    (2, 2, 16, 16, 256),   # full-config stem channels
+    (2, 2, 8, 8, 768),     # up-path concat channels (Cg=24)
+    (1, 2, 8, 8, 1536),    # biggest concat
+    (2, 2, 16, 16, 128),   # small config
+    (2, 2, 8, 8, 96),      # small concat (Cg=3 -> V=1 path)
+]
+
+
+@pytest.mark.parametrize("shape", GN_SHAPES)
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("film,silu", [(False, False), (False, True),
+                                       (True, True)])
+def test_gn_forward_parity(shape, dtype, film, silu):
+    B, F, H, W, C = shape
+    groups = min(32, C)
+    while C % groups:
+        groups -= 1
+    x, gamma, beta, fs, ft = _gn_case(B, F, H, W, C, dtype, film, silu)
+    got = hip_ops.joint_groupnorm(x, gamma, beta, groups, 1e-6, fs, ft, silu)
+    want = ref.joint_groupnorm(x.float(), gamma, beta, groups, 1e-6,
+                               None if fs is None else fs.float(),
+                               None if ft is None else ft.float(), silu)
+    tol = 5e-5 if dtype == torch.float32 else 2e-2
+    err = (got.float() - want).abs().max().item()
+    scale = want.abs().max().item() + 1e-6
+    assert err / scale < tol, f"rel err {err/scale:.2e}"
+
+
+@pytest.mark.parametrize("shape", GN_SHAPES[:3])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("film,silu", [(False, False), (True, True)])
+def test_gn_backward_parity(shape, dtype, film, silu):
+    B, F, H, W, C = shape
+    groups = min(32, C)
+    while C % groups:
+        groups -= 1
+    x, gamma, beta, fs, ft = _gn_case(B, F, H, W, C, dtype, film, silu)
+
+    def run(fn, xx, gm, bt, fss, ftt):
+        xx = xx.detach().clone().requires_grad_(True)
+        gm = gm.detach().clone().requires_grad_(True)
+        bt = bt.detach().clone().requires_grad_(True)
+        args = [xx, gm, bt]
+        if film:
+            fss = fss.detach().clone().requires_grad_(True)
+            ftt = ftt.detach().clone().requires_grad_(True)
+        y = fn(xx, gm, bt, groups, 1e-6, fss if film else None,
+               ftt if film else None, silu)
+        torch.manual_seed(0)
+        dy = torch.randn_like(y.float())
+        (y.float() * dy).sum().backward()
+        grads = [xx.grad, gm.grad, bt.grad]
+        if film:
+            grads += [fss.grad, ftt.grad]
+        return grads
+
+    got = run(hip_ops.joint_groupnorm, x, gamma, beta, fs, ft)
+    want = run(lambda *a: ref.joint_groupnorm(a[0].float(), *a[1:3], *a[3:]),
+               x.float(), gamma, beta,
+               None if fs is None else fs.float(),
+               None if ft is None else ft.float())
+    tol = 2e-4 if dtype == torch.float32 else 3e-2
+    names = ["dx", "dgamma", "dbeta", "dfscale", "dfshift"]
+    for n, gg, ww in zip(names, got, want):
+        err = (gg.float() - ww.float()).abs().max().item()
+        scale = ww.float().abs().max().item() + 1e-5
+        assert err / scale < tol, f"{n}: rel err {err/scale:.2e}"
+
+
+def test_pose_embedding_parity():
+    from novel_view_synthesis_3d_amd.data.synthetic import random_cameras
+    g = torch.Generator(device="cuda").manual_seed(0)
+    B, H = 3, 32
+    R1, t1, K = random_cameras(B, H, "cuda", g)
+    R2, t2, _ = random_cameras(B, H, "cuda", g)
+    R = torch.stack([R1, R2], 1)
+    t = torch.stack([t1, t2], 1)
+    mask = torch.tensor([1.0, 0.0, 1.0], device="cuda")
+    got = hip_ops.pose_embedding(R, t, K, mask, H, H, torch.float32)
+    want = ref.pose_embedding(R, t, K, mask, H, H, torch.float32)
+    assert got.shape == (B, 2, H, H, 144)
+    err = (got - want).abs().max().item()
+    assert err < 1e-4, err
+    # masked batch element must be exactly zero
+    assert got[1].abs().max().item() == 0.0
+
+
+def test_pose_embedding_bf16_out():
+    from novel_view_synthesis_3d_amd.data.synthetic import random_cameras
+    g = torch.Generator(device="cuda").manual_seed(1)
+    B, H = 2, 16
+    R1, t1, K = random_cameras(B, H, "cuda", g)
+    R2, t2, _ = random_cameras(B, H, "cuda", g)
+    R, t = torch.stack([R1, R2], 1), torch.stack([t1, t2], 1)
+    got = hip_ops.pose_embedding(R, t, K, None, H, H, torch.bfloat16)
+    want = ref.pose_embedding(R, t, K, None, H, H, torch.float32)
+    assert got.dtype == torch.bfloat16
+    err = (got.float() - want).abs().max().item()
+    assert err < 2e-2, err
+
+
+def test_fused_adam_matches_torch_adam():
+    from novel_view_synthesis_3d_amd.engine.optim import FusedAdam
+    torch.manual_seed(0)
+    shapes = [(1000,), (33,), (256, 129), (7, 3, 3, 5)]
+    params1 = [torch.randn(s, device="cuda") for s in shapes]
+    params2 = [p.detach().clone() for p in params1]
+    for p in params1 + params2:
+        p.requires_grad_(True)
+    opt1 = FusedAdam(params1, lr=1e-2, betas=(0.9, 0.99), eps=1e-8)
+    opt2 = torch.optim.Adam(params2, lr=1e-2, betas=(0.9, 0.99), eps=1e-8)
+    for step in range(5):
+        torch.manual_seed(step)
+        for p1, p2 in zip(params1, params2):
+            gr = torch.randn_like(p1)
+            p1.grad = gr.clone()
+            p2.grad = gr.clone()
+        opt1.step()
+        opt2.step()
+    for p1, p2 in zip(params1, params2):
+        err = (p1 - p2).abs().max().item()
+        assert err < 1e-5, err
+
+
+def test_native_extension_is_loaded():
+    """Guard against silent eager fallback on GPU boxes."""
+    import novel_view_synthesis_3d_amd.ops as ops
+    assert ops.hip_available()
+    x = torch.randn(1, 2, 4, 4, 32, device="cuda")
+    assert ops._use_hip(x, "joint_groupnorm")
