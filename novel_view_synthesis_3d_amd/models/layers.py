@@ -30,7 +30,9 @@ class FrameConv(nn.Module):
     """Per-frame 3x3 'SAME' conv — flax nn.Conv(kernel=(1,3,3), strides=(1,s,s))
     (/root/reference/model/xunet.py:81,85,199-202,229,276).
 
-    Weight stored (Cout, Cin, 3, 3); flax kernel layout is (1, 3, 3, Cin, Cout).
+    Weight stored (Cout, 3, 3, Cin) contiguous (OHWI = channels-last layout
+    of an OIHW conv weight, and the layout the CDNA4 implicit-GEMM conv
+    kernel consumes); flax kernel layout is (1, 3, 3, Cin, Cout).
     """
 
     flax_type = "Conv"
@@ -39,7 +41,7 @@ class FrameConv(nn.Module):
                  zero_init: bool = False):
         super().__init__()
         self.cin, self.cout, self.stride = cin, cout, stride
-        self.weight = nn.Parameter(torch.empty(cout, cin, 3, 3))
+        self.weight = nn.Parameter(torch.empty(cout, 3, 3, cin))
         self.bias = nn.Parameter(torch.zeros(cout))
         if zero_init:
             nn.init.zeros_(self.weight)
@@ -52,8 +54,8 @@ class FrameConv(nn.Module):
     def flax_leaves(self):
         return [
             ("kernel", self.weight,
-             lambda w: w.permute(2, 3, 1, 0).unsqueeze(0),      # -> (1,3,3,Cin,Cout)
-             lambda f: torch.as_tensor(f).squeeze(0).permute(3, 2, 0, 1)),
+             lambda w: w.permute(1, 2, 3, 0).unsqueeze(0),      # -> (1,3,3,Cin,Cout)
+             lambda f: torch.as_tensor(f).squeeze(0).permute(3, 0, 1, 2)),
             ("bias", self.bias, lambda b: b, lambda f: torch.as_tensor(f)),
         ]
 

@@ -34,17 +34,20 @@ def frame_conv3x3(x: torch.Tensor, weight: torch.Tensor,
 
     The reference's only conv type: kernel (1,3,3), stride (1,s,s)
     (/root/reference/model/xunet.py:81,85,199-202,229,276). `weight` is
-    (Cout, Cin, 3, 3) stored channels-last-compatible.
+    (Cout, 3, 3, Cin) contiguous (OHWI) — presented to conv2d as a
+    channels-last (Cout,Cin,3,3) view so MIOpen takes the NHWC path with no
+    transposes.
     """
     B, Fr, H, W, C = x.shape
     ph = _same_pad(H, 3, stride)
     pw = _same_pad(W, 3, stride)
     xf = x.reshape(B * Fr, H, W, C).permute(0, 3, 1, 2)  # NCHW view of NHWC data
+    w = weight.permute(0, 3, 1, 2)  # channels-last view of OHWI storage
     if ph != (1, 1) or pw != (1, 1) or stride != 1:
         xf = F.pad(xf, (pw[0], pw[1], ph[0], ph[1]))
-        y = F.conv2d(xf, weight, bias, stride=stride)
+        y = F.conv2d(xf, w, bias, stride=stride)
     else:
-        y = F.conv2d(xf, weight, bias, stride=1, padding=1)
+        y = F.conv2d(xf, w, bias, stride=1, padding=1)
     Ho, Wo = y.shape[-2], y.shape[-1]
     return y.permute(0, 2, 3, 1).reshape(B, Fr, Ho, Wo, -1)
 

@@ -1,0 +1,109 @@
+"""GPU integration tests: model forward/backward, trainer step, sampler graph.
+
+These run the REAL MI355X path: HIP kernels active (no eager fallback), bf16
+autocast, FusedAdam.
+"""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _mk_trainer(tmp_path, model="small", H=64, batch=2):
+    from novel_view_synthesis_3d_amd.config import TrainConfig, XUNetConfig
+    from novel_view_synthesis_3d_amd.engine.trainer import Trainer
+    cfg = TrainConfig()
+    cfg.data = "synthetic"
+    cfg.log_every = 10 ** 9
+    cfg.ckpt_folder = str(tmp_path / "ckpt")
+    return Trainer(None, train_batch_size=batch, train_lr=1e-3,
+                   train_num_steps=10 ** 9, img_sidelength=H,
+                   results_folder=str(tmp_path / "res"),
+                   model_cfg=XUNetConfig.named(model), train_cfg=cfg)
+
+
+def test_small_model_train_step_and_loss_decreases(tmp_path):
+    trainer = _mk_trainer(tmp_path)
+    first = [float(trainer.train_step().item()) for _ in range(3)]
+    for _ in range(30):
+        trainer.train_step()
+    last = [float(trainer.train_step().item()) for _ in range(3)]
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(first + last)))
+    assert sum(last) < sum(first), (first, last)
+
+
+def test_model_gpu_matches_cpu_eager_fp32(tmp_path):
+    """End-to-end numerics: full fp32 forward on GPU (HIP kernels) vs CPU
+    (eager oracle) on identical weights + inputs."""
+    from novel_view_synthesis_3d_amd.config import XUNetConfig
+    from novel_view_synthesis_3d_amd.models.xunet import XUNet
+    from tests.test_model import make_inputs
+
+    torch.manual_seed(0)
+    cfg = XUNetConfig(ch=32, ch_mult=(1, 2), emb_ch=32, num_res_blocks=1,
+                      attn_resolutions=(16,), dropout=0.0)
+    model = XUNet(cfg, img_sidelength=32)
+    with torch.no_grad():  # nonzero head so outputs are informative
+        model.Conv_1.weight.normal_(0, 0.05)
+    model.eval()
+    batch, _ = make_inputs(B=2, H=32)
+    out_cpu = model(batch, cond_mask=torch.ones(2))
+
+    model_gpu = model.cuda()
+    batch_gpu = {k: v.cuda() for k, v in batch.items()}
+    out_gpu = model_gpu(batch_gpu, cond_mask=torch.ones(2, device="cuda"))
+    err = (out_gpu.cpu() - out_cpu).abs().max().item()
+    scale = out_cpu.abs().max().item()
+    assert err < 5e-3 * max(scale, 1.0), (err, scale)
+
+
+def test_sampler_graph_capture_matches_eager(tmp_path):
+    from novel_view_synthesis_3d_amd.config import XUNetConfig
+    from novel_view_synthesis_3d_amd.data.synthetic import synthetic_batch
+    from novel_view_synthesis_3d_amd.diffusion.sampler import DDPMSampler
+    from novel_view_synthesis_3d_amd.models.xunet import XUNet
+
+    torch.manual_seed(0)
+    cfg = XUNetConfig(ch=32, ch_mult=(1, 2), emb_ch=32, num_res_blocks=1,
+                      attn_resolutions=(16,), dropout=0.0)
+    model = XUNet(cfg, img_sidelength=32).cuda()
+    with torch.no_grad():
+        model.Conv_1.weight.normal_(0, 0.05)
+    g = torch.Generator(device="cuda").manual_seed(0)
+    cond = synthetic_batch(2, 32, "cuda", g)
+    cond.pop("x_target")
+    z0 = torch.randn(2, 32, 32, 3, device="cuda", generator=g)
+
+    out_eager = DDPMSampler(model, num_steps=6, use_graph=False).sample(
+        cond, z_init=z0)
+    torch.manual_seed(7)
+    out_graph = DDPMSampler(model, num_steps=6, use_graph=True).sample(
+        cond, z_init=z0)
+    torch.manual_seed(7)
+    out_graph2 = DDPMSampler(model, num_steps=6, use_graph=True).sample(
+        cond, z_init=z0)
+    assert torch.isfinite(out_graph).all()
+    # graph path is deterministic under the same seed
+    assert torch.allclose(out_graph, out_graph2, atol=1e-5)
+    # and statistically consistent with the eager path (same clamp range)
+    assert out_eager.shape == out_graph.shape
+    assert out_graph.abs().max().item() < 50.0
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from novel_view_synthesis_3d_amd.engine import checkpoint as ckpt
+    trainer = _mk_trainer(tmp_path)
+    for _ in range(2):
+        trainer.train_step()
+    path = ckpt.save_checkpoint(str(tmp_path / "ckpt"), trainer.model,
+                                trainer.opt, 2)
+    trainer2 = _mk_trainer(tmp_path)
+    step = ckpt.load_checkpoint(path, trainer2.model, trainer2.opt,
+                                map_location="cuda")
+    assert step == 2
+    trainer2.train_step()
+    torch.cuda.synchronize()

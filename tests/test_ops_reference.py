@@ -48,14 +48,15 @@ def test_joint_groupnorm_film_silu_fusion():
 def test_frame_conv_same_padding_stride1():
     torch.manual_seed(2)
     x = torch.randn(1, 2, 8, 8, 4)
-    w = torch.randn(6, 4, 3, 3) * 0.1
+    w = torch.randn(6, 3, 3, 4) * 0.1  # OHWI layout
     b = torch.zeros(6)
     y = ref.frame_conv3x3(x, w, b)
     assert y.shape == (1, 2, 8, 8, 6)
     # compare against direct conv2d on each frame
     for f in range(2):
         xf = x[0, f].permute(2, 0, 1)[None]
-        yf = F.conv2d(xf, w, b, padding=1)[0].permute(1, 2, 0)
+        yf = F.conv2d(xf, w.permute(0, 3, 1, 2), b,
+                      padding=1)[0].permute(1, 2, 0)
         assert torch.allclose(y[0, f], yf, atol=1e-5)
 
 
@@ -64,7 +65,7 @@ def test_frame_conv_same_padding_strided(stride, H):
     """FLAX SAME semantics: out = ceil(H/s), asymmetric padding."""
     torch.manual_seed(3)
     x = torch.randn(1, 2, H, H, 3)
-    w = torch.randn(5, 3, 3, 3) * 0.1
+    w = torch.randn(5, 3, 3, 3) * 0.1  # OHWI
     y = ref.frame_conv3x3(x, w, None, stride=stride)
     assert y.shape[2] == -(-H // stride)
     assert y.shape == (1, 2, -(-H // stride), -(-H // stride), 5)
