@@ -41,7 +41,7 @@ def test_model_gpu_matches_cpu_eager_fp32(tmp_path):
     (eager oracle) on identical weights + inputs."""
     from novel_view_synthesis_3d_amd.config import XUNetConfig
     from novel_view_synthesis_3d_amd.models.xunet import XUNet
-    from tests.test_model import make_inputs
+    from test_model import make_inputs  # tests/ is on sys.path via conftest
 
     torch.manual_seed(0)
     cfg = XUNetConfig(ch=32, ch_mult=(1, 2), emb_ch=32, num_res_blocks=1,
