@@ -18,6 +18,8 @@ torch::Tensor rays_posenc(torch::Tensor R, torch::Tensor t, torch::Tensor Kinv,
 void fused_adam(torch::Tensor ptrs, torch::Tensor chunk_tensor,
                 torch::Tensor chunk_off, torch::Tensor numels,
                 double lr, double b1, double b2, double eps, int64_t step);
+torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -37,6 +39,7 @@ TORCH_LIBRARY(nvs3d, m) {
   m.def("fused_adam(Tensor ptrs, Tensor chunk_tensor, Tensor chunk_off, "
         "Tensor numels, float lr, float b1, float b2, float eps, "
         "int step) -> ()");
+  m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
@@ -44,4 +47,5 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("gn_bwd", gn_bwd);
   m.impl("rays_posenc", rays_posenc_py);
   m.impl("fused_adam", fused_adam);
+  m.impl("conv3x3_fwd", conv3x3_fwd);
 }

@@ -1,0 +1,270 @@
+// Implicit-GEMM 3x3 SAME stride-1 conv, NHWC bf16, MFMA — CDNA4 gfx950.
+//
+// the reference's only conv type (kernel (1,3,3),
+// /root/reference/model/xunet.py:81,85,229,276). GEMM view:
+//   out[M, N] = sum_k A[M, k] * B[k, N]
+//   M = B*F*H*W pixels, N = Cout, k = (plane dy*3+dx, ci), K = 9*Cin.
+// A is the im2col image — materialized on the fly into LDS with boundary
+// masking (zero padding); B is the weight, whose OHWI storage
+// (Cout, 3, 3, Cin) is ALREADY (n, k) row-major, so both stage coalesced.
+//
+// Structure (cdna_hip_programming.md §5 ladder step 2-3): 128x128 tile,
+// BK=64, 4 waves (2x2 of 64x64), mfma_f32_16x16x32_bf16 with 4x4
+// accumulator fragments per wave, double-buffered LDS with the
+// ((row&7)<<4) byte-XOR swizzle (§6 G4) on both tiles, register-staged
+// A/B (boundary masks need per-lane predication, so no glds), epilogue
+// through LDS for coalesced bf16 stores, bias fused.
+//
+// The same kernel computes dgrad: conv3x3(dy, w~) with
+// w~[ci, ey, ex, co] = w[co, 2-ey, 2-ex, ci] (host-side transform).
+// Constraints: Cin % 64 == 0, Cout % 128 == 0 (dispatch falls back to the
+// MIOpen path otherwise — stem 3ch / posemb 144ch / head 3ch convs).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int BM = 128;
+constexpr int BN = 128;
+constexpr int BK = 64;
+constexpr int THREADS = 256;
+// LDS tile: [rows][BK] bf16, byte-swizzled. row stride = BK*2 = 128B.
+__device__ __forceinline__ int swz(int row, int k_elem) {
+  int byte = row * (BK * 2) + k_elem * 2;
+  return byte ^ ((row & 7) << 4);
+}
+
+struct ConvShape {
+  int IMG;       // B*F images
+  int H, W;      // spatial
+  int Cin, Cout;
+  int M;         // IMG*H*W
+  int ksteps;    // 9*Cin / BK
+  int steps_per_plane;  // Cin / BK
+};
+
+__global__ __launch_bounds__(THREADS)
+void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
+                   const bf16* __restrict__ w,   // (Cout, 9*Cin) rows=co
+                   const float* __restrict__ bias,  // (Cout,) or null
+                   bf16* __restrict__ out,       // (IMG,H,W,Cout)
+                   ConvShape s, int nblocks_m) {
+  // XCD-aware block swizzle (bijective form, §5.5 T1): consecutive
+  // swizzled ids walk M-blocks within one N-column so each XCD's L2 keeps
+  // the weight panel.
+  int bid = blockIdx.x;
+  const int nwg = nblocks_m * (s.Cout / BN);
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int bm = bid % nblocks_m;
+  const int bn = bid / nblocks_m;
+  const int m0 = bm * BM;
+  const int n0 = bn * BN;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* ldsA = reinterpret_cast<bf16*>(smem);               // 2 x 16KB
+  bf16* ldsB = reinterpret_cast<bf16*>(smem + 2 * BM * BK * 2);  // 2 x 16KB
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;   // 0..1: M half
+  const int wc = wave & 1;    // 0..1: N half
+
+  // --- staging helpers -------------------------------------------------
+  // A tile: BM rows x BK k. 8 bf16 per 16B pack; BK/8 = 8 packs per row;
+  // BM*8 = 1024 packs / 256 threads = 4 per thread.
+  // Pack p: row = p >> 3, kp = p & 7.
+  // B tile: BN rows x BK; same geometry.
+  // Row -> pixel decode is K-step-invariant: hoist the divisions here.
+  int rowh[4], roww[4], rowok[4];
+  long rowbase[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int row = (tid + it * THREADS) >> 3;
+    const int m = m0 + row;
+    const int wpix = m % s.W;
+    const int hpix = (m / s.W) % s.H;
+    const int img = m / (s.W * s.H);
+    rowh[it] = hpix;
+    roww[it] = wpix;
+    rowok[it] = m < s.M;
+    rowbase[it] = ((long)img * s.H) * s.W * s.Cin;
+  }
+
+  float acc[4][4][4];  // [mi][nj][reg] f32x4 fragments
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  Pack<bf16, 8> regA[4], regB[4];
+
+  auto stage_load = [&](int kstep) {
+    const int plane = kstep / s.steps_per_plane;
+    const int ci0 = (kstep % s.steps_per_plane) * BK;
+    const int dy = plane / 3 - 1;
+    const int dx = plane % 3 - 1;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int p = tid + it * THREADS;
+      const int row = p >> 3;
+      const int kp = p & 7;
+      const int hh = rowh[it] + dy;
+      const int ww = roww[it] + dx;
+      const bool valid = rowok[it] & (hh >= 0) & (hh < s.H) & (ww >= 0)
+                         & (ww < s.W);
+      if (valid) {
+        const long src = rowbase[it] + ((long)hh * s.W + ww) * s.Cin;
+        regA[it] = pload<bf16, 8>(x + src + ci0 + kp * 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) regA[it].v[j] = __float2bfloat16(0.f);
+      }
+      // B: row n (= co), k contiguous in OHWI storage
+      const int co = n0 + row;
+      regB[it] = pload<bf16, 8>(w + (long)co * (9 * s.Cin)
+                                + kstep * BK + kp * 8);
+    }
+  };
+
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int p = tid + it * THREADS;
+      const int row = p >> 3;
+      const int kp = p & 7;
+      *reinterpret_cast<Pack<bf16, 8>*>(
+          reinterpret_cast<char*>(ldsA) + buf * BM * BK * 2
+          + swz(row, kp * 8)) = regA[it];
+      *reinterpret_cast<Pack<bf16, 8>*>(
+          reinterpret_cast<char*>(ldsB) + buf * BN * BK * 2
+          + swz(row, kp * 8)) = regB[it];
+    }
+  };
+
+  auto compute = [&](int buf) {
+    const char* baseA = reinterpret_cast<const char*>(ldsA) + buf * BM * BK * 2;
+    const char* baseB = reinterpret_cast<const char*>(ldsB) + buf * BN * BK * 2;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bf[4];
+      const int kbase = kk * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = wr * 64 + i * 16 + (lane & 15);
+        af[i] = *reinterpret_cast<const bf16x8*>(baseA + swz(row, kbase));
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int row = wc * 64 + j * 16 + (lane & 15);
+        bf[j] = *reinterpret_cast<const bf16x8*>(baseB + swz(row, kbase));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          *reinterpret_cast<f32x4*>(acc[i][j]) =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[i], bf[j], *reinterpret_cast<f32x4*>(acc[i][j]),
+                  0, 0, 0);
+        }
+    }
+  };
+
+  // --- main loop: simple 2-buffer pipeline, one barrier per K-step -----
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  for (int t = 0; t < s.ksteps; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < s.ksteps) stage_load(t + 1);
+    compute(cur);
+    if (t + 1 < s.ksteps) stage_write(cur ^ 1);
+    __syncthreads();
+  }
+
+  // --- epilogue: acc -> (bias add) -> bf16 via LDS -> coalesced stores --
+  bf16* ldsC = reinterpret_cast<bf16*>(smem);  // [BM][BN] bf16 = 32 KB
+  __syncthreads();  // done with A/B buffers
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int col = wc * 64 + j * 16 + (lane & 15);
+    const float bj = bias != nullptr ? bias[n0 + col] : 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int rbase = wr * 64 + i * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        ldsC[(rbase + r) * BN + col] = __float2bfloat16(acc[i][j][r] + bj);
+      }
+    }
+  }
+  __syncthreads();
+  // store: 128 rows x 16 packs(16B) = 2048 packs / 256 threads = 8 each
+#pragma unroll
+  for (int it = 0; it < 8; ++it) {
+    const int p = tid + it * THREADS;
+    const int row = p >> 4;
+    const int cp = p & 15;
+    const int m = m0 + row;
+    if (m < s.M) {
+      Pack<bf16, 8> v = *reinterpret_cast<Pack<bf16, 8>*>(
+          ldsC + row * BN + cp * 8);
+      pstore<bf16, 8>(out + (long)m * s.Cout + n0 + cp * 8, v);
+    }
+  }
+}
+
+}  // namespace
+
+torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias) {
+  // x: (B,F,H,W,Cin) or (IMG,H,W,Cin) bf16 contiguous; w: (Cout,3,3,Cin)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16);
+  auto xs = x.sizes();
+  const int nd = x.dim();
+  TORCH_CHECK(nd == 4 || nd == 5);
+  ConvShape s;
+  s.IMG = nd == 5 ? xs[0] * xs[1] : xs[0];
+  s.H = xs[nd - 3]; s.W = xs[nd - 2]; s.Cin = xs[nd - 1];
+  s.Cout = w.size(0);
+  TORCH_CHECK(w.size(1) == 3 && w.size(2) == 3 && w.size(3) == s.Cin);
+  TORCH_CHECK(s.Cin % BK == 0, "Cin must be multiple of 64");
+  TORCH_CHECK(s.Cout % BN == 0, "Cout must be multiple of 128");
+  s.M = s.IMG * s.H * s.W;
+  s.steps_per_plane = s.Cin / BK;
+  s.ksteps = 9 * s.steps_per_plane;
+
+  std::vector<int64_t> oshape(xs.begin(), xs.end());
+  oshape[nd - 1] = s.Cout;
+  auto out = torch::empty(oshape, x.options());
+
+  torch::Tensor biasf;
+  if (bias.has_value()) biasf = bias->to(torch::kFloat).contiguous();
+
+  const int nblocks_m = (s.M + BM - 1) / BM;
+  const int grid = nblocks_m * (s.Cout / BN);
+  const size_t lds = 2 * (BM + BN) * BK * 2;  // 64 KB
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(conv3x3_igemm, dim3(grid), dim3(THREADS), lds, stream,
+      reinterpret_cast<const bf16*>(x.data_ptr()),
+      reinterpret_cast<const bf16*>(w.data_ptr()),
+      bias.has_value() ? biasf.data_ptr<float>() : nullptr,
+      reinterpret_cast<bf16*>(out.data_ptr()), s, nblocks_m);
+  return out;
+}

@@ -16,11 +16,21 @@ from novel_view_synthesis_3d_amd.ops.build import built_path
 _SO = built_path()
 if not os.path.exists(_SO):
     raise ImportError(f"nvs3d_hip extension not built (expected {_SO})")
-torch.ops.load_library(_SO)
+try:  # the build step may already have registered the library in-process
+    torch.ops.nvs3d.gn_fwd
+except (AttributeError, RuntimeError):
+    torch.ops.load_library(_SO)
 _OPS = torch.ops.nvs3d
 
 # ops with a HIP implementation (consulted by ops/__init__.py dispatch)
-HAS = {"joint_groupnorm", "pose_embedding"}
+HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3"}
+
+
+def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
+    """The implicit-GEMM kernel covers the FLOP-dominant convs (stride 1,
+    Cin%64==0, Cout%128==0); stem (3ch), pose-emb (144ch), head (3ch) and
+    small-config convs take the MIOpen path."""
+    return stride == 1 and cin % 64 == 0 and cout % 128 == 0
 
 
 # ---------------------------------------------------------------------------
@@ -115,3 +125,74 @@ class AdamPlan:
 def fused_adam_step(plan: AdamPlan, lr, b1, b2, eps, step: int) -> None:
     _OPS.fused_adam(plan.ptrs, plan.chunk_tensor, plan.chunk_off, plan.numels,
                     lr, b1, b2, eps, step)
+
+
+# ---------------------------------------------------------------------------
+# Implicit-GEMM 3x3 conv (K1): fwd + dgrad on the MFMA kernel, wgrad via
+# MIOpen convolution_backward (library path for the long-K reduction GEMM).
+# ---------------------------------------------------------------------------
+
+class _FrameConv3x3(torch.autograd.Function):
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x, weight, bias):
+        x = x.contiguous()
+        weight = weight.contiguous()
+        y = _OPS.conv3x3_fwd(x, weight, bias)
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        cout, _, _, cin = w.shape
+        need_x, need_w, need_b = (ctx.needs_input_grad[0],
+                                  ctx.needs_input_grad[1], ctx.has_bias)
+        dx = dw = db = None
+        if need_x:
+            # dgrad = conv3x3(dy, w~), w~[ci,ey,ex,co] = w[co,2-ey,2-ex,ci]
+            wt = torch.flip(w, dims=(1, 2)).permute(3, 1, 2, 0).contiguous()
+            if conv_shapes_supported(cout, cin, 1):
+                dx = _OPS.conv3x3_fwd(dy, wt, None)
+            else:
+                dx = _miopen_conv(dy, wt, None)
+        if need_w or need_b:
+            dw4, db = _miopen_wgrad(x, w, dy, ctx.has_bias)
+            if need_w:
+                dw = dw4
+        return dx, dw, db
+
+
+def _nchw_view(x5):
+    """(IMG,H,W,C) or (B,F,H,W,C) contiguous -> NCHW channels_last view."""
+    if x5.dim() == 5:
+        B, F, H, W, C = x5.shape
+        x5 = x5.reshape(B * F, H, W, C)
+    return x5.permute(0, 3, 1, 2)
+
+
+def _miopen_conv(x, w, bias):
+    import torch.nn.functional as Fn
+    shape = x.shape
+    y = Fn.conv2d(_nchw_view(x), w.permute(0, 3, 1, 2), bias, padding=1)
+    out = y.permute(0, 2, 3, 1)
+    return out.reshape(*shape[:-1], w.shape[0]).contiguous()
+
+
+def _miopen_wgrad(x, w, dy, has_bias):
+    xg = _nchw_view(x)
+    dyg = _nchw_view(dy)
+    wg = w.permute(0, 3, 1, 2)  # (Cout,Cin,3,3) channels_last view
+    _, dw, db = torch.ops.aten.convolution_backward(
+        dyg, xg, wg, [w.shape[0]] if has_bias else None,
+        [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+        [False, True, has_bias])
+    dw_ohwi = dw.permute(0, 2, 3, 1).contiguous()
+    return dw_ohwi, db
+
+
+def frame_conv3x3(x, weight, bias, stride: int = 1):
+    return _FrameConv3x3.apply(x, weight, bias)

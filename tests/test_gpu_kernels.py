@@ -162,3 +162,55 @@ def test_native_extension_is_loaded():
     assert ops.hip_available()
     x = torch.randn(1, 2, 4, 4, 32, device="cuda")
     assert ops._use_hip(x, "joint_groupnorm")
+
+
+CONV_SHAPES = [
+    (2, 2, 16, 16, 256, 256),
+    (1, 2, 16, 16, 768, 128),
+    (1, 2, 8, 8, 1024, 512),
+    (1, 2, 32, 32, 64, 128),   # small-ish, W=32
+]
+
+
+@pytest.mark.parametrize("shape", CONV_SHAPES)
+def test_conv3x3_forward_parity(shape):
+    B, F, H, W, Cin, Cout = shape
+    g = torch.Generator(device="cuda").manual_seed(0)
+    x = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                    dtype=torch.bfloat16)
+    w = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                    dtype=torch.bfloat16) * (1.0 / (3 * Cin ** 0.5))
+    b = torch.randn(Cout, device="cuda", generator=g) * 0.1
+    got = torch.ops.nvs3d.conv3x3_fwd(x, w, b)
+    want = ref.frame_conv3x3(x.float(), w.float(), b.float())
+    err = (got.float() - want).abs().max().item()
+    scale = want.abs().max().item()
+    assert err < 3e-2 * max(scale, 1.0), (err, scale)
+
+
+def test_conv3x3_autograd_parity():
+    B, F, H, W, Cin, Cout = 1, 2, 16, 16, 256, 128
+    g = torch.Generator(device="cuda").manual_seed(1)
+    x0 = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    w0 = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * (1.0 / (3 * Cin ** 0.5))
+    b0 = torch.randn(Cout, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.1
+    dy = torch.randn(B, F, H, W, Cout, device="cuda", generator=g)
+
+    def run(fn, dtype):
+        x = x0.detach().to(dtype).requires_grad_(True)
+        w = w0.detach().to(dtype).requires_grad_(True)
+        b = b0.detach().to(dtype).requires_grad_(True)
+        y = fn(x, w, b)
+        (y.float() * dy).sum().backward()
+        return y.float(), x.grad.float(), w.grad.float(), b.grad.float()
+
+    got = run(lambda x, w, b: hip_ops.frame_conv3x3(x, w, b), torch.bfloat16)
+    want = run(lambda x, w, b: ref.frame_conv3x3(x, w, b), torch.float32)
+    names = ["y", "dx", "dw", "db"]
+    for n, gg, ww in zip(names, got, want):
+        err = (gg - ww).abs().max().item()
+        scale = ww.abs().max().item() + 1e-6
+        assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
