@@ -92,7 +92,11 @@ def joint_groupnorm(x, gamma, beta, groups: int, eps: float = 1e-6,
 
 def attention(q, k, v):
     if _use_hip(q, "attention"):
-        return _HIP_MOD.attention(q, k, v)
+        B, L, h, d = q.shape
+        bf16_path = (q.dtype == torch.bfloat16
+                     or torch.is_autocast_enabled())
+        if bf16_path and L % 64 == 0 and d in (16, 32, 64, 128, 256):
+            return _HIP_MOD.attention(q, k, v)
     return ref.attention(q, k, v)
 
 

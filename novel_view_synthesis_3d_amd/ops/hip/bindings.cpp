@@ -20,6 +20,8 @@ void fused_adam(torch::Tensor ptrs, torch::Tensor chunk_tensor,
                 double lr, double b1, double b2, double eps, int64_t step);
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -40,6 +42,7 @@ TORCH_LIBRARY(nvs3d, m) {
         "Tensor numels, float lr, float b1, float b2, float eps, "
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
+  m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
@@ -48,4 +51,5 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("rays_posenc", rays_posenc_py);
   m.impl("fused_adam", fused_adam);
   m.impl("conv3x3_fwd", conv3x3_fwd);
+  m.impl("attn_fwd", attn_fwd);
 }

@@ -23,7 +23,7 @@ except (AttributeError, RuntimeError):
 _OPS = torch.ops.nvs3d
 
 # ops with a HIP implementation (consulted by ops/__init__.py dispatch)
-HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3"}
+HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3", "attention"}
 
 
 def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
@@ -196,3 +196,47 @@ def _miopen_wgrad(x, w, dy, has_bias):
 
 def frame_conv3x3(x, weight, bias, stride: int = 1):
     return _FrameConv3x3.apply(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# Flash MFMA attention (K7): HIP forward (saves logsumexp), GEMM-recompute
+# backward (rocBLAS batched matmuls — the "plain library GEMM" path).
+# ---------------------------------------------------------------------------
+
+class _Attention(torch.autograd.Function):
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, q, k, v):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        out, lse = _OPS.attn_fwd(q, k, v)
+        ctx.save_for_backward(q, k, v, out, lse)
+        return out
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        B, L, H, D = q.shape
+        scale = 1.0 / (D ** 0.5)
+        # (B,H,L,D) views
+        qt = q.permute(0, 2, 1, 3)
+        kt = k.permute(0, 2, 1, 3)
+        vt = v.permute(0, 2, 1, 3)
+        ot = o.permute(0, 2, 1, 3).float()
+        dot = do.permute(0, 2, 1, 3).float()
+        s = torch.matmul(qt, kt.transpose(-1, -2)).float() * scale
+        p = torch.exp(s - lse.permute(0, 2, 1)[..., None])  # (B,H,L,Lk)
+        pb = p.to(q.dtype)
+        dv = torch.matmul(pb.transpose(-1, -2), dot.to(q.dtype))
+        dp = torch.matmul(dot.to(q.dtype), vt.transpose(-1, -2)).float()
+        delta = (dot * ot).sum(-1, keepdim=True)
+        ds = (p * (dp - delta)).to(q.dtype)
+        dq = torch.matmul(ds, kt) * scale
+        dk = torch.matmul(ds.transpose(-1, -2), qt) * scale
+        return (dq.permute(0, 2, 1, 3).to(q.dtype),
+                dk.permute(0, 2, 1, 3).to(q.dtype),
+                dv.permute(0, 2, 1, 3).to(q.dtype))
+
+
+def attention(q, k, v):
+    return _Attention.apply(q, k, v)
