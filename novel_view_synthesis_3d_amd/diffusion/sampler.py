@@ -88,7 +88,7 @@ class DDPMSampler:
         return c2, mask
 
     def _step(self, z: torch.Tensor, idx: torch.Tensor, cond2, mask, tab,
-              noise: Optional[torch.Tensor] = None) -> None:
+              pose_cache=None) -> None:
         """One in-place ancestral step; everything device-side."""
         B = z.shape[0]
         logsnr = tab["logsnr"].index_select(0, idx).expand(2 * B)
@@ -97,9 +97,9 @@ class DDPMSampler:
         batch["logsnr"] = logsnr
         if z.is_cuda:
             with torch.autocast("cuda", dtype=torch.bfloat16):
-                out = self.model(batch, mask)
+                out = self.model(batch, mask, pose_cache=pose_cache)
         else:
-            out = self.model(batch, mask)
+            out = self.model(batch, mask, pose_cache=pose_cache)
         out = out.to(torch.float32)
         eps = (1.0 + self.w) * out[:B] - self.w * out[B:]
 
@@ -110,8 +110,7 @@ class DDPMSampler:
         if self.clip_denoised:
             x0 = x0.clamp(-1.0, 1.0)
         mean = c["mean_c1"] * x0 + c["mean_c2"] * z
-        if noise is None:
-            noise = torch.randn_like(z)
+        noise = torch.randn_like(z)
         z.copy_(mean + c["sigma"] * noise)
         idx.add_(1)
 
@@ -136,17 +135,27 @@ class DDPMSampler:
             tab = self._step_tables(device)
             idx = torch.zeros(1, dtype=torch.long, device=device)
 
+            # Pose conditioning is step-invariant: compute once, outside any
+            # graph capture (fixes both the 2x redundant work per step and
+            # capture-unsupported ops in the camera math).
+            cp = self.model.ConditioningProcessor_0
+            if device.type == "cuda":
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    pose_cache = cp.pose_features(cond2, mask)
+            else:
+                pose_cache = cp.pose_features(cond2, mask)
+
             if self.use_graph and device.type == "cuda":
-                self._run_graphed(z, idx, cond2, mask, tab)
+                self._run_graphed(z, idx, cond2, mask, tab, pose_cache)
             else:
                 for _ in range(self.num_steps):
-                    self._step(z, idx, cond2, mask, tab)
+                    self._step(z, idx, cond2, mask, tab, pose_cache)
             return z
         finally:
             self.model.train(was_training)
 
     # -----------------------------------------------------------------
-    def _run_graphed(self, z, idx, cond2, mask, tab) -> None:
+    def _run_graphed(self, z, idx, cond2, mask, tab, pose_cache) -> None:
         """Capture one sampler step as a hipGraph and replay it num_steps
         times. Per-step state (z, idx) lives in the captured buffers; RNG is
         graph-safe (torch captures the philox offset)."""
@@ -154,7 +163,7 @@ class DDPMSampler:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):  # warmup, required before capture
             for _ in range(2):
-                self._step(z, idx, cond2, mask, tab)
+                self._step(z, idx, cond2, mask, tab, pose_cache)
         torch.cuda.current_stream().wait_stream(s)
         done_warmup = int(idx.item())
 
@@ -162,6 +171,6 @@ class DDPMSampler:
         with torch.cuda.graph(graph):
             # capture RECORDS the step without executing it; z/idx advance
             # only on replay
-            self._step(z, idx, cond2, mask, tab)
+            self._step(z, idx, cond2, mask, tab, pose_cache)
         for _ in range(self.num_steps - done_warmup):
             graph.replay()

@@ -17,6 +17,31 @@ from __future__ import annotations
 import torch
 
 
+def inv3x3(K: torch.Tensor) -> torch.Tensor:
+    """Closed-form batched 3x3 inverse (adjugate/det). Unlike
+    torch.linalg.inv this is pure elementwise math: hipGraph-capture-safe
+    and faster for the tiny camera matrices."""
+    K = K.to(torch.float32)
+    a, b, c = K[..., 0, 0], K[..., 0, 1], K[..., 0, 2]
+    d, e, f = K[..., 1, 0], K[..., 1, 1], K[..., 1, 2]
+    g, h, i = K[..., 2, 0], K[..., 2, 1], K[..., 2, 2]
+    A = e * i - f * h
+    B = c * h - b * i
+    C = b * f - c * e
+    D = f * g - d * i
+    E = a * i - c * g
+    F = c * d - a * f
+    G = d * h - e * g
+    Hc = b * g - a * h
+    I = a * e - b * d
+    det = a * A + b * D + c * G
+    inv = torch.stack([
+        torch.stack([A, B, C], dim=-1),
+        torch.stack([D, E, F], dim=-1),
+        torch.stack([G, Hc, I], dim=-1)], dim=-2)
+    return inv / det[..., None, None]
+
+
 def camera_rays(R: torch.Tensor, t: torch.Tensor, K: torch.Tensor,
                 H: int, W: int):
     """Compute per-pixel ray origins and unit directions.
@@ -36,7 +61,7 @@ def camera_rays(R: torch.Tensor, t: torch.Tensor, K: torch.Tensor,
         torch.arange(W, device=device, dtype=dtype) + 0.5,
         indexing="ij")
     px = torch.stack([u, v, torch.ones_like(u)], dim=-1)        # (H, W, 3)
-    Kinv = torch.linalg.inv(K.to(torch.float32)).to(dtype)      # (B, 3, 3)
+    Kinv = inv3x3(K).to(dtype)                                  # (B, 3, 3)
     # dir_cam[b,h,w,:] = Kinv[b] @ px[h,w,:]
     dir_cam = torch.einsum("bij,hwj->bhwi", Kinv, px)
     dir_world = torch.einsum("bij,bhwj->bhwi", R, dir_cam)
