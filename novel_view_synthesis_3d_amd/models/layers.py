@@ -138,10 +138,10 @@ class JointGroupNorm(nn.Module):
         self.scale = nn.Parameter(torch.ones(channels))
         self.bias = nn.Parameter(torch.zeros(channels))
 
-    def forward(self, x: torch.Tensor, film_scale=None, film_shift=None,
+    def forward(self, x: torch.Tensor, film=None,
                 silu: bool = False) -> torch.Tensor:
         return ops.joint_groupnorm(x, self.scale, self.bias, self.groups,
-                                   self.EPS, film_scale, film_shift, silu)
+                                   self.EPS, film, silu)
 
     def flax_leaves(self):
         return [
@@ -154,8 +154,11 @@ class JointGroupNorm(nn.Module):
 
 class FiLM(nn.Module):
     """Feature-wise linear modulation (/root/reference/model/xunet.py:54-61):
-    Dense(2*features)(silu(emb)) -> scale, shift. The modulation itself is
-    fused into the GroupNorm kernel (ops.joint_groupnorm)."""
+    Dense(2*features)(silu(emb)) -> packed (.., 2*features) scale|shift. The
+    modulation itself is fused into the GroupNorm kernel
+    (ops.joint_groupnorm), which consumes the packed tensor strided — no
+    split/copy. The caller passes emb ALREADY silu'd (XUNet caches silu(emb)
+    once per level instead of once per block)."""
 
     flax_type = "FiLM"
 
@@ -164,9 +167,8 @@ class FiLM(nn.Module):
         self.features = features
         self.Dense_0 = Dense(emb_ch, 2 * features)
 
-    def scale_shift(self, emb: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        p = self.Dense_0(F.silu(emb))
-        return p[..., :self.features], p[..., self.features:]
+    def packed(self, emb_silu: torch.Tensor) -> torch.Tensor:
+        return self.Dense_0(emb_silu)
 
 
 # ---------------------------------------------------------------------------
@@ -205,7 +207,8 @@ class ResnetBlock(nn.Module):
         else:
             self.Dense_0 = None
 
-    def forward(self, h_in: torch.Tensor, emb: torch.Tensor) -> torch.Tensor:
+    def forward(self, h_in: torch.Tensor, emb_silu: torch.Tensor
+                ) -> torch.Tensor:
         h = self.GroupNorm_0(h_in, silu=True)
         if self.resample == "up":
             h = ops.nearest_upsample2x(h)
@@ -214,8 +217,8 @@ class ResnetBlock(nn.Module):
             h = ops.avgpool_downsample2x(h)
             h_in = ops.avgpool_downsample2x(h_in)
         h = self.Conv_0(h)
-        scale, shift = self.FiLM_0.scale_shift(emb)
-        h = self.GroupNorm_1(h, film_scale=scale, film_shift=shift, silu=True)
+        film = self.FiLM_0.packed(emb_silu)
+        h = self.GroupNorm_1(h, film=film, silu=True)
         if self.dropout_rate > 0.0:
             h = F.dropout(h, self.dropout_rate, training=self.training)
         h = self.Conv_1(h)

@@ -205,9 +205,16 @@ class XUNet(nn.Module):
         logsnr_emb, pose_embs = self.ConditioningProcessor_0(
             batch, cond_mask, pose_cache=pose_cache)
 
+        _emb_cache: Dict[int, torch.Tensor] = {}
+
         def emb_at(level: int) -> torch.Tensor:
-            # (B,1,1,1,E) + (B,2,H',W',E)  (xunet.py:233)
-            return logsnr_emb[:, None, None, None, :] + pose_embs[level]
+            # silu((B,1,1,1,E) + (B,2,H',W',E))  (xunet.py:233,59) — the
+            # silu belongs to FiLM's input; cached per level since every
+            # block of a level consumes the identical tensor.
+            if level not in _emb_cache:
+                e = logsnr_emb[:, None, None, None, :] + pose_embs[level]
+                _emb_cache[level] = F.silu(e)
+            return _emb_cache[level]
 
         h = torch.stack([batch["x"], batch["z"]], dim=1)  # (B,2,H,W,3)
         h = self.Conv_0(h)

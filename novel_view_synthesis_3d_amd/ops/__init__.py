@@ -82,12 +82,11 @@ def frame_conv3x3(x, weight, bias, stride: int = 1):
 
 
 def joint_groupnorm(x, gamma, beta, groups: int, eps: float = 1e-6,
-                    film_scale=None, film_shift=None, silu: bool = False):
+                    film=None, silu: bool = False):
     if _use_hip(x, "joint_groupnorm"):
         return _HIP_MOD.joint_groupnorm(x, gamma, beta, groups, eps,
-                                        film_scale, film_shift, silu)
-    return ref.joint_groupnorm(x, gamma, beta, groups, eps,
-                               film_scale, film_shift, silu)
+                                        film, silu)
+    return ref.joint_groupnorm(x, gamma, beta, groups, eps, film, silu)
 
 
 def attention(q, k, v):

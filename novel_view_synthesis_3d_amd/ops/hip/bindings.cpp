@@ -3,13 +3,11 @@
 
 std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
-                                  c10::optional<torch::Tensor> fscale,
-                                  c10::optional<torch::Tensor> fshift,
+                                  c10::optional<torch::Tensor> film,
                                   int64_t groups, double eps, bool silu);
 std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor beta,
-                                  c10::optional<torch::Tensor> fscale,
-                                  c10::optional<torch::Tensor> fshift,
+                                  c10::optional<torch::Tensor> film,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   int64_t groups, bool silu);
 torch::Tensor rays_posenc(torch::Tensor R, torch::Tensor t, torch::Tensor Kinv,
@@ -31,10 +29,10 @@ torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
 }
 
 TORCH_LIBRARY(nvs3d, m) {
-  m.def("gn_fwd(Tensor x, Tensor gamma, Tensor beta, Tensor? fscale, "
-        "Tensor? fshift, int groups, float eps, bool silu) -> Tensor[]");
+  m.def("gn_fwd(Tensor x, Tensor gamma, Tensor beta, Tensor? film, "
+        "int groups, float eps, bool silu) -> Tensor[]");
   m.def("gn_bwd(Tensor dy, Tensor x, Tensor gamma, Tensor beta, "
-        "Tensor? fscale, Tensor? fshift, Tensor mean, Tensor rstd, "
+        "Tensor? film, Tensor mean, Tensor rstd, "
         "int groups, bool silu) -> Tensor[]");
   m.def("rays_posenc(Tensor R, Tensor t, Tensor Kinv, Tensor? mask, "
         "int H, int W, Tensor dtype_like) -> Tensor");

@@ -94,8 +94,7 @@ __global__ void gn_fwd_apply(const T* __restrict__ x,
                              const float* __restrict__ partials,
                              const float* __restrict__ gamma,
                              const float* __restrict__ beta,
-                             const T* __restrict__ fscale,
-                             const T* __restrict__ fshift,
+                             const T* __restrict__ film,  // (..,2C): scale|shift
                              T* __restrict__ y,
                              float* __restrict__ mean_out,   // (B,G)
                              float* __restrict__ rstd_out,   // (B,G)
@@ -138,11 +137,14 @@ __global__ void gn_fwd_apply(const T* __restrict__ x,
   const int r1 = min(s.R, r0 + rc);
   const size_t base = (size_t)b * s.R * s.C;
 
+  const size_t fbase = (size_t)b * s.R * 2 * s.C;
   for (int row = r0 + ri; row < r1; row += rowsPerIter) {
     const size_t off = base + (size_t)row * s.C + c0;
+    const size_t foff = fbase + (size_t)row * 2 * s.C + c0;
     Pack<T, V> px = pload<T, V>(x + off);
     Pack<T, V> ps, pt;
-    if (FILM) { ps = pload<T, V>(fscale + off); pt = pload<T, V>(fshift + off); }
+    if (FILM) { ps = pload<T, V>(film + foff);
+                pt = pload<T, V>(film + foff + s.C); }
     Pack<T, V> po;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -164,15 +166,13 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
                                 const T* __restrict__ x,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                const T* __restrict__ fscale,
-                                const T* __restrict__ fshift,
+                                const T* __restrict__ film,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ rstd,
                                 float* __restrict__ partials,   // (B,G,P,2)
                                 float* __restrict__ dgamma,     // (C,) zeroed
                                 float* __restrict__ dbeta,      // (C,) zeroed
-                                T* __restrict__ dfscale,
-                                T* __restrict__ dfshift,
+                                T* __restrict__ dfilm,
                                 GnShape s) {
   __shared__ float ls1[MAX_GROUPS], ls2[MAX_GROUPS];
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -206,12 +206,15 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
   const size_t base = (size_t)b * s.R * s.C;
   float s1 = 0.f, s2 = 0.f;
 
+  const size_t fbase = (size_t)b * s.R * 2 * s.C;
   for (int row = r0 + ri; row < r1; row += rowsPerIter) {
     const size_t off = base + (size_t)row * s.C + c0;
+    const size_t foff = fbase + (size_t)row * 2 * s.C + c0;
     Pack<T, V> px = pload<T, V>(x + off);
     Pack<T, V> pdy = pload<T, V>(dy + off);
     Pack<T, V> ps, pt;
-    if (FILM) { ps = pload<T, V>(fscale + off); pt = pload<T, V>(fshift + off); }
+    if (FILM) { ps = pload<T, V>(film + foff);
+                pt = pload<T, V>(film + foff + s.C); }
     Pack<T, V> pds, pdt;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -236,8 +239,8 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
       s2 += dxh * xhat;
     }
     if (FILM) {
-      pstore<T, V>(dfscale + off, pds);
-      pstore<T, V>(dfshift + off, pdt);
+      pstore<T, V>(dfilm + foff, pds);
+      pstore<T, V>(dfilm + foff + s.C, pdt);
     }
   }
   atomicAdd(&ls1[g], s1);
@@ -267,8 +270,7 @@ __global__ void gn_bwd_apply(const T* __restrict__ dy,
                              const T* __restrict__ x,
                              const float* __restrict__ gamma,
                              const float* __restrict__ beta,
-                             const T* __restrict__ fscale,
-                             const T* __restrict__ fshift,
+                             const T* __restrict__ film,
                              const float* __restrict__ mean,
                              const float* __restrict__ rstd,
                              const float* __restrict__ partials,
@@ -307,12 +309,15 @@ __global__ void gn_bwd_apply(const T* __restrict__ dy,
   const int r1 = min(s.R, r0 + rc);
   const size_t base = (size_t)b * s.R * s.C;
 
+  const size_t fbase = (size_t)b * s.R * 2 * s.C;
   for (int row = r0 + ri; row < r1; row += rowsPerIter) {
     const size_t off = base + (size_t)row * s.C + c0;
+    const size_t foff = fbase + (size_t)row * 2 * s.C + c0;
     Pack<T, V> px = pload<T, V>(x + off);
     Pack<T, V> pdy = pload<T, V>(dy + off);
     Pack<T, V> ps, pt;
-    if (FILM) { ps = pload<T, V>(fscale + off); pt = pload<T, V>(fshift + off); }
+    if (FILM) { ps = pload<T, V>(film + foff);
+                pt = pload<T, V>(film + foff + s.C); }
     Pack<T, V> pdx;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -389,8 +394,7 @@ bool pick_block(GnShape& s, int V) {
 // ---------------------------------------------------------------------------
 std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
-                                  c10::optional<torch::Tensor> fscale,
-                                  c10::optional<torch::Tensor> fshift,
+                                  c10::optional<torch::Tensor> film_in,
                                   int64_t groups, double eps, bool silu) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous CUDA");
   TORCH_CHECK(x.dim() == 5, "x must be (B,F,H,W,C)");
@@ -398,11 +402,11 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
   const long R = x.size(1) * x.size(2) * x.size(3);
   const long C = x.size(4);
   TORCH_CHECK(C % groups == 0 && groups <= MAX_GROUPS);
-  const bool film = fscale.has_value();
+  const bool film = film_in.has_value();
   if (film) {
-    TORCH_CHECK(fscale->is_contiguous() && fshift->is_contiguous());
-    TORCH_CHECK(fscale->sizes() == x.sizes(), "film scale must match x");
-    TORCH_CHECK(fscale->scalar_type() == x.scalar_type());
+    TORCH_CHECK(film_in->is_contiguous());
+    TORCH_CHECK(film_in->size(4) == 2 * C, "film must be (..,2C) scale|shift");
+    TORCH_CHECK(film_in->scalar_type() == x.scalar_type());
   }
 
   GnShape s = make_shape(B, R, C, groups);
@@ -436,8 +440,7 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
               reinterpret_cast<const T*>(x.data_ptr()),
               partials.data_ptr<float>(),
               gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
-              film ? reinterpret_cast<const T*>(fscale->data_ptr()) : nullptr,
-              film ? reinterpret_cast<const T*>(fshift->data_ptr()) : nullptr,
+              film ? reinterpret_cast<const T*>(film_in->data_ptr()) : nullptr,
               reinterpret_cast<T*>(y.data_ptr()),
               mean.data_ptr<float>(), rstd.data_ptr<float>(),
               (float)eps, s);
@@ -450,8 +453,7 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
 
 std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor beta,
-                                  c10::optional<torch::Tensor> fscale,
-                                  c10::optional<torch::Tensor> fshift,
+                                  c10::optional<torch::Tensor> film_in,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   int64_t groups, bool silu) {
   TORCH_CHECK(dy.is_cuda() && x.is_contiguous());
@@ -459,7 +461,7 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
   const long B = x.size(0);
   const long R = x.size(1) * x.size(2) * x.size(3);
   const long C = x.size(4);
-  const bool film = fscale.has_value();
+  const bool film = film_in.has_value();
 
   GnShape s = make_shape(B, R, C, groups);
   const int elem = x.scalar_type() == torch::kFloat ? 4 : 2;
@@ -471,10 +473,9 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
   auto dgamma = torch::zeros({C}, opts);
   auto dbeta = torch::zeros({C}, opts);
   auto dx = torch::empty_like(x);
-  torch::Tensor dfscale, dfshift;
+  torch::Tensor dfilm;
   if (film) {
-    dfscale = torch::empty_like(*fscale);
-    dfshift = torch::empty_like(*fshift);
+    dfilm = torch::empty_like(*film_in);
   }
 
   auto stream = at::hip::getCurrentHIPStream();
@@ -495,21 +496,18 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
               reinterpret_cast<const T*>(dyc.data_ptr()),
               reinterpret_cast<const T*>(x.data_ptr()),
               gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
-              film ? reinterpret_cast<const T*>(fscale->data_ptr()) : nullptr,
-              film ? reinterpret_cast<const T*>(fshift->data_ptr()) : nullptr,
+              film ? reinterpret_cast<const T*>(film_in->data_ptr()) : nullptr,
               mean.data_ptr<float>(), rstd.data_ptr<float>(),
               partials.data_ptr<float>(),
               dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-              film ? reinterpret_cast<T*>(dfscale.data_ptr()) : nullptr,
-              film ? reinterpret_cast<T*>(dfshift.data_ptr()) : nullptr,
+              film ? reinterpret_cast<T*>(dfilm.data_ptr()) : nullptr,
               s);
           hipLaunchKernelGGL((gn_bwd_apply<T, V, FILM, SILU>), grid, block, 0,
               stream,
               reinterpret_cast<const T*>(dyc.data_ptr()),
               reinterpret_cast<const T*>(x.data_ptr()),
               gammaf.data_ptr<float>(), betaf.data_ptr<float>(),
-              film ? reinterpret_cast<const T*>(fscale->data_ptr()) : nullptr,
-              film ? reinterpret_cast<const T*>(fshift->data_ptr()) : nullptr,
+              film ? reinterpret_cast<const T*>(film_in->data_ptr()) : nullptr,
               mean.data_ptr<float>(), rstd.data_ptr<float>(),
               partials.data_ptr<float>(),
               reinterpret_cast<T*>(dx.data_ptr()), s);
@@ -517,6 +515,6 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
       }));
     }));
   });
-  if (film) return {dx, dgamma, dbeta, dfscale, dfshift};
+  if (film) return {dx, dgamma, dbeta, dfilm};
   return {dx, dgamma, dbeta};
 }

@@ -39,41 +39,37 @@ def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
 
 class _GnFused(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, fscale, fshift, groups, eps, silu):
+    def forward(ctx, x, gamma, beta, film, groups, eps, silu):
         x = x.contiguous()
-        film = fscale is not None
-        if film:
-            fscale = fscale.to(x.dtype).contiguous()
-            fshift = fshift.to(x.dtype).contiguous()
-        y, mean, rstd = _OPS.gn_fwd(x, gamma, beta, fscale, fshift,
-                                    groups, eps, silu)
+        has_film = film is not None
+        if has_film:
+            film = film.to(x.dtype).contiguous()
+        y, mean, rstd = _OPS.gn_fwd(x, gamma, beta, film, groups, eps, silu)
         ctx.save_for_backward(x, gamma, beta, mean, rstd,
-                              *( (fscale, fshift) if film else () ))
-        ctx.groups, ctx.silu, ctx.film = groups, silu, film
+                              *((film,) if has_film else ()))
+        ctx.groups, ctx.silu, ctx.film = groups, silu, has_film
         return y
 
     @staticmethod
     def backward(ctx, dy):
         if ctx.film:
-            x, gamma, beta, mean, rstd, fscale, fshift = ctx.saved_tensors
+            x, gamma, beta, mean, rstd, film = ctx.saved_tensors
         else:
             x, gamma, beta, mean, rstd = ctx.saved_tensors
-            fscale = fshift = None
-        outs = _OPS.gn_bwd(dy, x, gamma, beta, fscale, fshift, mean, rstd,
+            film = None
+        outs = _OPS.gn_bwd(dy, x, gamma, beta, film, mean, rstd,
                            ctx.groups, ctx.silu)
         if ctx.film:
-            dx, dgamma, dbeta, dfs, dft = outs
+            dx, dgamma, dbeta, dfilm = outs
         else:
             dx, dgamma, dbeta = outs
-            dfs = dft = None
+            dfilm = None
         return (dx, dgamma.to(gamma.dtype), dbeta.to(beta.dtype),
-                dfs, dft, None, None, None)
+                dfilm, None, None, None)
 
 
-def joint_groupnorm(x, gamma, beta, groups, eps=1e-6,
-                    film_scale=None, film_shift=None, silu=False):
-    return _GnFused.apply(x, gamma, beta, film_scale, film_shift,
-                          groups, eps, silu)
+def joint_groupnorm(x, gamma, beta, groups, eps=1e-6, film=None, silu=False):
+    return _GnFused.apply(x, gamma, beta, film, groups, eps, silu)
 
 
 # ---------------------------------------------------------------------------

@@ -54,15 +54,15 @@ def frame_conv3x3(x: torch.Tensor, weight: torch.Tensor,
 
 def joint_groupnorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                     groups: int, eps: float = 1e-6,
-                    film_scale: Optional[torch.Tensor] = None,
-                    film_shift: Optional[torch.Tensor] = None,
+                    film: Optional[torch.Tensor] = None,
                     silu: bool = False) -> torch.Tensor:
     """K3(+K5+K4): GroupNorm with statistics jointly over BOTH frames and all
     spatial positions per (batch, group) — the reference's frame-axis GroupNorm
     (/root/reference/model/xunet.py:46-52; flax GroupNorm reduces over all
     non-batch axes). Optionally fused FiLM modulate (xunet.py:54-61) and SiLU.
 
-    x: (B, F, H, W, C); gamma/beta: (C,); film_scale/shift: broadcastable to x.
+    x: (B, F, H, W, C); gamma/beta: (C,); film: (B, F, H, W, 2C) packed
+    scale|shift (one Dense output, consumed strided — no split copies).
     Stats in fp32; output in x.dtype.
     """
     B, Fr, H, W, C = x.shape
@@ -73,8 +73,10 @@ def joint_groupnorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
     h = (xf - mean) * torch.rsqrt(var + eps)
     h = h.reshape(B, Fr, H, W, C)
     h = h * gamma.to(torch.float32) + beta.to(torch.float32)
-    if film_scale is not None:
-        h = h * (1.0 + film_scale.to(torch.float32)) + film_shift.to(torch.float32)
+    if film is not None:
+        scale = film[..., :C].to(torch.float32)
+        shift = film[..., C:].to(torch.float32)
+        h = h * (1.0 + scale) + shift
     if silu:
         h = F.silu(h)
     return h.to(x.dtype)

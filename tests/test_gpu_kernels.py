@@ -50,10 +50,11 @@ def test_gn_forward_parity(shape, dtype, film, silu):
     while C % groups:
         groups -= 1
     x, gamma, beta, fs, ft = _gn_case(B, F, H, W, C, dtype, film, silu)
-    got = hip_ops.joint_groupnorm(x, gamma, beta, groups, 1e-6, fs, ft, silu)
+    packed = None if fs is None else torch.cat([fs, ft], dim=-1)
+    got = hip_ops.joint_groupnorm(x, gamma, beta, groups, 1e-6, packed, silu)
     want = ref.joint_groupnorm(x.float(), gamma, beta, groups, 1e-6,
-                               None if fs is None else fs.float(),
-                               None if ft is None else ft.float(), silu)
+                               None if packed is None else packed.float(),
+                               silu)
     tol = 5e-5 if dtype == torch.float32 else 2e-2
     err = (got.float() - want).abs().max().item()
     scale = want.abs().max().item() + 1e-6
@@ -69,32 +70,29 @@ def test_gn_backward_parity(shape, dtype, film, silu):
     while C % groups:
         groups -= 1
     x, gamma, beta, fs, ft = _gn_case(B, F, H, W, C, dtype, film, silu)
+    packed0 = None if fs is None else torch.cat([fs, ft], dim=-1)
 
-    def run(fn, xx, gm, bt, fss, ftt):
+    def run(fn, xx, gm, bt, fpk):
         xx = xx.detach().clone().requires_grad_(True)
         gm = gm.detach().clone().requires_grad_(True)
         bt = bt.detach().clone().requires_grad_(True)
-        args = [xx, gm, bt]
         if film:
-            fss = fss.detach().clone().requires_grad_(True)
-            ftt = ftt.detach().clone().requires_grad_(True)
-        y = fn(xx, gm, bt, groups, 1e-6, fss if film else None,
-               ftt if film else None, silu)
+            fpk = fpk.detach().clone().requires_grad_(True)
+        y = fn(xx, gm, bt, groups, 1e-6, fpk if film else None, silu)
         torch.manual_seed(0)
         dy = torch.randn_like(y.float())
         (y.float() * dy).sum().backward()
         grads = [xx.grad, gm.grad, bt.grad]
         if film:
-            grads += [fss.grad, ftt.grad]
+            grads += [fpk.grad]
         return grads
 
-    got = run(hip_ops.joint_groupnorm, x, gamma, beta, fs, ft)
-    want = run(lambda *a: ref.joint_groupnorm(a[0].float(), *a[1:3], *a[3:]),
+    got = run(hip_ops.joint_groupnorm, x, gamma, beta, packed0)
+    want = run(lambda *a: ref.joint_groupnorm(a[0].float(), *a[1:]),
                x.float(), gamma, beta,
-               None if fs is None else fs.float(),
-               None if ft is None else ft.float())
+               None if packed0 is None else packed0.float())
     tol = 2e-4 if dtype == torch.float32 else 3e-2
-    names = ["dx", "dgamma", "dbeta", "dfscale", "dfshift"]
+    names = ["dx", "dgamma", "dbeta", "dfilm"]
     for n, gg, ww in zip(names, got, want):
         err = (gg.float() - ww.float()).abs().max().item()
         scale = ww.float().abs().max().item() + 1e-5

@@ -38,8 +38,9 @@ def test_joint_groupnorm_film_silu_fusion():
     beta = torch.randn(8) * 0.1
     scale = torch.randn(2, 2, 4, 4, 8) * 0.1
     shift = torch.randn(2, 2, 4, 4, 8) * 0.1
-    fused = ref.joint_groupnorm(x, gamma, beta, 4, film_scale=scale,
-                                film_shift=shift, silu=True)
+    fused = ref.joint_groupnorm(x, gamma, beta, 4,
+                                film=torch.cat([scale, shift], dim=-1),
+                                silu=True)
     base = ref.joint_groupnorm(x, gamma, beta, 4)
     manual = F.silu(base * (1 + scale) + shift)
     assert torch.allclose(fused, manual, atol=1e-5)

@@ -85,7 +85,8 @@ def bench_gn():
         fs = torch.randn_like(x) * 0.1
         ft = torch.randn_like(x) * 0.1
         nbytes = x.numel() * 2
-        t = timeit(lambda: torch.ops.nvs3d.gn_fwd(x, gm, bt, fs, ft, 32,
+        film = torch.cat([fs, ft], dim=-1).contiguous()
+        t = timeit(lambda: torch.ops.nvs3d.gn_fwd(x, gm, bt, film, 32,
                                                   1e-6, True))
         # fwd traffic: read x twice + fs + ft + write y = 5 passes
         rec = {"op": "gn_fwd_film_silu", "shape": [B, F, H, W, C],
