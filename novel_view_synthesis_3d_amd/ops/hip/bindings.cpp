@@ -20,6 +20,7 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
+torch::Tensor im2col3x3(torch::Tensor x);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -41,6 +42,7 @@ TORCH_LIBRARY(nvs3d, m) {
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
+  m.def("im2col3x3(Tensor x) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
@@ -50,4 +52,5 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("fused_adam", fused_adam);
   m.impl("conv3x3_fwd", conv3x3_fwd);
   m.impl("attn_fwd", attn_fwd);
+  m.impl("im2col3x3", im2col3x3);
 }

@@ -1,0 +1,68 @@
+// im2col for the 3x3 SAME stride-1 conv — CDNA4 gfx950.
+//
+// Materializes A[m][k] (M = IMG*H*W pixels, k = (plane, ci), K = 9*Cin,
+// zero-padded borders) so the conv WGRAD becomes one hipBLASLt GEMM
+// A^T (9Cin x M) @ dy (M x Cout) — the long-K reduction that MIOpen's wrw
+// kernels handle poorly on gfx950 (measured 4x slower than the GEMM peak
+// path; see profiles/). Thread mapping: one 16-byte pack per thread-step,
+// coalesced on both sides.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+
+__global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
+                                 bf16* __restrict__ out,      // (M, 9*C)
+                                 int IMG, int H, int W, int C) {
+  const long M = (long)IMG * H * W;
+  const int packs_per_plane = C / 8;
+  const long total = M * 9 * packs_per_plane;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int cp = (int)(i % packs_per_plane);
+    long rest = i / packs_per_plane;
+    const int plane = (int)(rest % 9);
+    const long m = rest / 9;
+    const int wpix = (int)(m % W);
+    const int hpix = (int)((m / W) % H);
+    const int img = (int)(m / ((long)W * H));
+    const int hh = hpix + plane / 3 - 1;
+    const int ww = wpix + plane % 3 - 1;
+    Pack<bf16, 8> v;
+    if (hh >= 0 && hh < H && ww >= 0 && ww < W) {
+      v = pload<bf16, 8>(x + (((long)img * H + hh) * W + ww) * C + cp * 8);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = __float2bfloat16(0.f);
+    }
+    pstore<bf16, 8>(out + m * (9L * C) + (long)plane * C + cp * 8, v);
+  }
+}
+
+}  // namespace
+
+torch::Tensor im2col3x3(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+              && x.scalar_type() == torch::kBFloat16);
+  const int nd = x.dim();
+  TORCH_CHECK(nd == 4 || nd == 5);
+  const int IMG = nd == 5 ? x.size(0) * x.size(1) : x.size(0);
+  const int H = x.size(nd - 3), W = x.size(nd - 2), C = x.size(nd - 1);
+  TORCH_CHECK(C % 8 == 0, "Cin must be a multiple of 8");
+  const long M = (long)IMG * H * W;
+  auto out = torch::empty({M, 9L * C}, x.options());
+  const int block = 256;
+  const long total = M * 9 * (C / 8);
+  const int grid = (int)std::min<long>((total + block - 1) / block, 16384);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(im2col3x3_kernel, dim3(grid), dim3(block), 0, stream,
+      reinterpret_cast<const bf16*>(x.data_ptr()),
+      reinterpret_cast<bf16*>(out.data_ptr()), IMG, H, W, C);
+  return out;
+}

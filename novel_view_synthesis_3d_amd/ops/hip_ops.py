@@ -157,7 +157,7 @@ class _FrameConv3x3(torch.autograd.Function):
             else:
                 dx = _miopen_conv(dy, wt, None)
         if need_w or need_b:
-            dw4, db = _miopen_wgrad(x, w, dy, ctx.has_bias)
+            dw4, db = _wgrad_im2col_gemm(x, w, dy, ctx.has_bias)
             if need_w:
                 dw = dw4
         return dx, dw, db
@@ -177,6 +177,20 @@ def _miopen_conv(x, w, bias):
     y = Fn.conv2d(_nchw_view(x), w.permute(0, 3, 1, 2), bias, padding=1)
     out = y.permute(0, 2, 3, 1)
     return out.reshape(*shape[:-1], w.shape[0]).contiguous()
+
+
+def _wgrad_im2col_gemm(x, w, dy, has_bias):
+    """Conv wgrad as im2col + ONE hipBLASLt GEMM: dW = A^T @ dy
+    (A = im2col(x), M x 9Cin). The long-K reduction runs at the GEMM peak
+    instead of MIOpen's slow wrw kernels; dbias is a column reduce."""
+    cout, _, _, cin = w.shape
+    A = _OPS.im2col3x3(x)                       # (M, 9Cin) bf16
+    dyf = dy.reshape(-1, cout)                  # (M, Cout)
+    dwf = torch.matmul(A.transpose(0, 1), dyf)  # (9Cin, Cout), fp32 accum
+    # (3,3,Cin,Cout) -> OHWI (Cout,3,3,Cin)
+    dw = dwf.reshape(3, 3, cin, cout).permute(3, 0, 1, 2).contiguous()
+    db = dyf.sum(dim=0, dtype=torch.float32).to(dy.dtype) if has_bias else None
+    return dw, db
 
 
 def _miopen_wgrad(x, w, dy, has_bias):
