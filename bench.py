@@ -33,6 +33,8 @@ def main() -> None:
     ap.add_argument("--model", default="full")
     ap.add_argument("--sidelength", type=int, default=128)
     ap.add_argument("--amp", default="bf16", choices=["bf16", "off"])
+    ap.add_argument("--graph", default="on", choices=["on", "off"],
+                    help="hipGraph-capture the train step (single-node)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -45,6 +47,7 @@ def main() -> None:
     train_cfg.data = "synthetic"
     train_cfg.amp = args.amp
     train_cfg.seed = 1234
+    train_cfg.use_graph = (args.graph == "on" and use_cuda)
 
     trainer = Trainer(None,
                       train_batch_size=args.batch,

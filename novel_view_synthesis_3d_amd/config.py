@@ -88,6 +88,7 @@ class TrainConfig:
     adam_betas: Tuple[float, float] = (0.9, 0.999)
     adam_eps: float = 1e-8
     bucket_mb: float = 40.0            # DDP gradient bucket size (xGMI sizing)
+    use_graph: bool = False            # hipGraph-capture the training step
     data: str = "auto"                 # "auto" | "synthetic" | "srn"
     num_workers: int = 4
     resume: Optional[str] = None       # checkpoint path to resume from

@@ -179,6 +179,8 @@ class Trainer:
     def train_step(self, raw: Optional[Dict[str, torch.Tensor]] = None
                    ) -> torch.Tensor:
         """One optimization step; returns the (local) loss tensor."""
+        if self.cfg.use_graph and self.device.type == "cuda":
+            return self._train_step_graphed(raw)
         if raw is None:
             raw = self.next_batch()
         batch, cond_mask, noise = self.prepare_model_inputs(raw)
@@ -192,6 +194,93 @@ class Trainer:
         self.ddp.finish()
         self.opt.step()
         return loss.detach()
+
+    # -- hipGraph-captured training step --------------------------------
+    # Forward noising + CFG mask + bf16 forward + loss + backward are
+    # captured as ONE hipGraph (kills per-kernel launch overhead and host
+    # gaps). Gradients land in stable p.grad storage; the bucketed RCCL
+    # all-reduce and the fused Adam update run after each replay (RCCL
+    # collectives are kept outside the capture).
+
+    def _graph_body(self):
+        if self.data_mode == "synthetic":
+            raw = synthetic_batch(self.cfg.train_batch_size,
+                                  self.cfg.img_sidelength, device=self.device)
+        else:
+            raw = self._static_raw
+        B = raw["x"].shape[0]
+        t = torch.randint(0, self.schedule.timesteps, (B,),
+                          device=self.device)
+        z, noise, logsnr = q_sample(raw["x_target"], t, self.schedule)
+        batch = {"x": raw["x"], "z": z, "logsnr": logsnr,
+                 "R1": raw["R1"], "t1": raw["t1"],
+                 "R2": raw["R2"], "t2": raw["t2"], "K": raw["K"]}
+        cond_mask = (torch.rand(B, device=self.device)
+                     > self.cfg.cond_drop_prob).to(torch.float32)
+        for p in self.model.parameters():  # zero stable grad storage
+            if p.grad is not None:
+                p.grad.zero_()
+        with self._autocast():
+            out = self.model(batch, cond_mask)
+        loss = self.compute_loss(out, noise)
+        with self.ddp.no_sync():  # collectives stay outside the graph
+            loss.backward()
+        return loss
+
+    def _init_graph(self):
+        if self.data_mode != "synthetic":
+            raw = self.next_batch()
+            self._static_raw = {k: v.clone() for k, v in raw.items()}
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):  # warmup allocates grads + caches plans
+                loss = self._graph_body()
+                self.opt.step()
+        torch.cuda.current_stream().wait_stream(s)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._static_loss = self._graph_body()
+
+    def _train_step_graphed(self, raw):
+        if not hasattr(self, "_graph"):
+            try:
+                self._init_graph()
+            except Exception as e:
+                import warnings
+                warnings.warn(f"hipGraph capture failed ({e}); falling back "
+                              f"to eager stepping")
+                self.cfg.use_graph = False
+                torch.cuda.synchronize()
+                return self.train_step(raw)
+        if self.data_mode != "synthetic":
+            if raw is None:
+                raw = self.next_batch()
+            for k, v in raw.items():
+                self._static_raw[k].copy_(v, non_blocking=True)
+        self._graph.replay()
+        if self.ddp.enabled:
+            self._allreduce_grads_post()
+        self.opt.step()
+        return self._static_loss.detach()
+
+    def _allreduce_grads_post(self):
+        """Bucketed all-reduce of the grads a replay produced (the in-graph
+        backward cannot launch RCCL ops)."""
+        for b in self.ddp.buckets:
+            b.ready = len(b.params)
+            b.launched = False
+        for b in self.ddp.buckets:
+            self.ddp._launch(b)
+        inv = 1.0 / self.ddp.world
+        for b in self.ddp.buckets:
+            b.handle.wait()
+            b.buffer.mul_(inv)
+            off = 0
+            for p in b.params:
+                n = p.numel()
+                p.grad.reshape(-1).copy_(b.buffer[off:off + n])
+                off += n
 
     def train(self) -> None:
         cfg = self.cfg
