@@ -31,6 +31,16 @@ using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
+// address-space casts for global_load_lds (via integer, the sanctioned way)
+using as1_cvp = const __attribute__((address_space(1))) void*;
+using as3_vp = __attribute__((address_space(3))) void*;
+__device__ __forceinline__ as1_cvp as_global(const void* p) {
+  return (as1_cvp)(unsigned long long)(uintptr_t)p;
+}
+__device__ __forceinline__ as3_vp as_shared(void* p) {
+  return (as3_vp)(unsigned int)(uintptr_t)p;
+}
+
 constexpr int BM = 128;
 constexpr int BN = 128;
 constexpr int BK = 64;
@@ -54,6 +64,7 @@ __global__ __launch_bounds__(THREADS)
 void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
                    const bf16* __restrict__ w,   // (Cout, 9*Cin) rows=co
                    const float* __restrict__ bias,  // (Cout,) or null
+                   const bf16* __restrict__ zbuf,   // 128B of zeros (padding)
                    bf16* __restrict__ out,       // (IMG,H,W,Cout)
                    ConvShape s, int nblocks_m) {
   // XCD-aware block swizzle (bijective form, §5.5 T1): consecutive
@@ -81,25 +92,30 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
   const int wr = wave >> 1;   // 0..1: M half
   const int wc = wave & 1;    // 0..1: N half
 
-  // --- staging helpers -------------------------------------------------
-  // A tile: BM rows x BK k. 8 bf16 per 16B pack; BK/8 = 8 packs per row;
-  // BM*8 = 1024 packs / 256 threads = 4 per thread.
-  // Pack p: row = p >> 3, kp = p & 7.
-  // B tile: BN rows x BK; same geometry.
-  // Row -> pixel decode is K-step-invariant: hoist the divisions here.
-  int rowh[4], roww[4], rowok[4];
-  long rowbase[4];
+  // --- glds staging ---------------------------------------------------
+  // A and B tiles stream global->LDS via global_load_lds (16B/lane): the
+  // LDS image is lane-linear, so the XOR swizzle moves to the per-lane
+  // SOURCE address (rule 21), and boundary zero-fill is done by pointing
+  // out-of-bounds lanes at a 128B zero buffer. One glds instruction fills
+  // 1 KiB: 16 per 16KB tile = 4 per wave; slot s of this wave covers LDS
+  // bytes [(wave*4+s)*1024, +1024).
+  // Lane's element under the swizzle: row = o/128 (XOR keeps the row),
+  // k-byte = (o ^ ((row&7)<<4)) % 128, with o = slot*1024 + lane*16.
+  int a_h[4], a_w[4], a_ok[4], a_kp[4], b_kp[4], b_co[4];
+  long a_base[4];
 #pragma unroll
-  for (int it = 0; it < 4; ++it) {
-    const int row = (tid + it * THREADS) >> 3;
+  for (int slot = 0; slot < 4; ++slot) {
+    const int o = (wave * 4 + slot) * 1024 + lane * 16;
+    const int row = o >> 7;
+    const int kb = (o ^ ((row & 7) << 4)) & 127;
+    a_kp[slot] = kb >> 4;
+    b_kp[slot] = kb >> 4;
+    b_co[slot] = n0 + row;
     const int m = m0 + row;
-    const int wpix = m % s.W;
-    const int hpix = (m / s.W) % s.H;
-    const int img = m / (s.W * s.H);
-    rowh[it] = hpix;
-    roww[it] = wpix;
-    rowok[it] = m < s.M;
-    rowbase[it] = ((long)img * s.H) * s.W * s.Cin;
+    a_w[slot] = m % s.W;
+    a_h[slot] = (m / s.W) % s.H;
+    a_ok[slot] = m < s.M;
+    a_base[slot] = ((long)(m / (s.W * s.H)) * s.H) * s.W * s.Cin;
   }
 
   float acc[4][4][4];  // [mi][nj][reg] f32x4 fragments
@@ -110,48 +126,31 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  Pack<bf16, 8> regA[4], regB[4];
-
-  auto stage_load = [&](int kstep) {
+  auto stage_glds = [&](int kstep, int buf) {
     const int plane = kstep / s.steps_per_plane;
     const int ci0 = (kstep % s.steps_per_plane) * BK;
     const int dy = plane / 3 - 1;
     const int dx = plane % 3 - 1;
+    char* baseA = reinterpret_cast<char*>(ldsA) + buf * BM * BK * 2
+                  + wave * 4 * 1024;
+    char* baseB = reinterpret_cast<char*>(ldsB) + buf * BN * BK * 2
+                  + wave * 4 * 1024;
 #pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      const int p = tid + it * THREADS;
-      const int row = p >> 3;
-      const int kp = p & 7;
-      const int hh = rowh[it] + dy;
-      const int ww = roww[it] + dx;
-      const bool valid = rowok[it] & (hh >= 0) & (hh < s.H) & (ww >= 0)
+    for (int slot = 0; slot < 4; ++slot) {
+      const int hh = a_h[slot] + dy;
+      const int ww = a_w[slot] + dx;
+      const bool valid = a_ok[slot] & (hh >= 0) & (hh < s.H) & (ww >= 0)
                          & (ww < s.W);
-      if (valid) {
-        const long src = rowbase[it] + ((long)hh * s.W + ww) * s.Cin;
-        regA[it] = pload<bf16, 8>(x + src + ci0 + kp * 8);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) regA[it].v[j] = __float2bfloat16(0.f);
-      }
-      // B: row n (= co), k contiguous in OHWI storage
-      const int co = n0 + row;
-      regB[it] = pload<bf16, 8>(w + (long)co * (9 * s.Cin)
-                                + kstep * BK + kp * 8);
-    }
-  };
-
-  auto stage_write = [&](int buf) {
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      const int p = tid + it * THREADS;
-      const int row = p >> 3;
-      const int kp = p & 7;
-      *reinterpret_cast<Pack<bf16, 8>*>(
-          reinterpret_cast<char*>(ldsA) + buf * BM * BK * 2
-          + swz(row, kp * 8)) = regA[it];
-      *reinterpret_cast<Pack<bf16, 8>*>(
-          reinterpret_cast<char*>(ldsB) + buf * BN * BK * 2
-          + swz(row, kp * 8)) = regB[it];
+      const bf16* srcA = valid
+          ? x + a_base[slot] + ((long)hh * s.W + ww) * s.Cin + ci0
+              + a_kp[slot] * 8
+          : zbuf;
+      __builtin_amdgcn_global_load_lds(as_global(srcA),
+          as_shared(baseA + slot * 1024), 16, 0, 0);
+      const bf16* srcB = w + (long)b_co[slot] * (9 * s.Cin) + kstep * BK
+                         + b_kp[slot] * 8;
+      __builtin_amdgcn_global_load_lds(as_global(srcB),
+          as_shared(baseB + slot * 1024), 16, 0, 0);
     }
   };
 
@@ -184,16 +183,14 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
     }
   };
 
-  // --- main loop: simple 2-buffer pipeline, one barrier per K-step -----
-  stage_load(0);
-  stage_write(0);
-  __syncthreads();
+  // --- main loop: glds double-buffer, one barrier per K-step ----------
+  stage_glds(0, 0);
+  __syncthreads();  // drains the glds (vmcnt(0) inside the barrier)
   for (int t = 0; t < s.ksteps; ++t) {
     const int cur = t & 1;
-    if (t + 1 < s.ksteps) stage_load(t + 1);
+    if (t + 1 < s.ksteps) stage_glds(t + 1, cur ^ 1);
     compute(cur);
-    if (t + 1 < s.ksteps) stage_write(cur ^ 1);
-    __syncthreads();
+    __syncthreads();  // glds for t+1 landed; LDS reads of buf cur done
   }
 
   // --- epilogue: acc -> (bias add) -> bf16 via LDS -> coalesced stores --
@@ -256,6 +253,10 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
 
   torch::Tensor biasf;
   if (bias.has_value()) biasf = bias->to(torch::kFloat).contiguous();
+  static torch::Tensor zbuf;  // 128B zero pad source for OOB glds lanes
+  if (!zbuf.defined() || zbuf.device() != x.device()) {
+    zbuf = torch::zeros({64}, x.options());
+  }
 
   const int nblocks_m = (s.M + BM - 1) / BM;
   const int grid = nblocks_m * (s.Cout / BN);
@@ -265,6 +266,7 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
       reinterpret_cast<const bf16*>(x.data_ptr()),
       reinterpret_cast<const bf16*>(w.data_ptr()),
       bias.has_value() ? biasf.data_ptr<float>() : nullptr,
+      reinterpret_cast<const bf16*>(zbuf.data_ptr()),
       reinterpret_cast<bf16*>(out.data_ptr()), s, nblocks_m);
   return out;
 }
