@@ -20,7 +20,8 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
-torch::Tensor im2col3x3(torch::Tensor x);
+torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
+                        c10::optional<torch::Tensor> out_buf);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -42,7 +43,7 @@ TORCH_LIBRARY(nvs3d, m) {
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
-  m.def("im2col3x3(Tensor x) -> Tensor");
+  m.def("im2col3x3(Tensor x, int m0, int m1, Tensor? out_buf) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {

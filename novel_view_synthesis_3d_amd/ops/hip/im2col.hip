@@ -17,18 +17,18 @@ namespace {
 using bf16 = __hip_bfloat16;
 
 __global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
-                                 bf16* __restrict__ out,      // (M, 9*C)
-                                 int IMG, int H, int W, int C) {
-  const long M = (long)IMG * H * W;
+                                 bf16* __restrict__ out,      // (m1-m0, 9*C)
+                                 int IMG, int H, int W, int C,
+                                 long m0, long m1) {
   const int packs_per_plane = C / 8;
-  const long total = M * 9 * packs_per_plane;
+  const long total = (m1 - m0) * 9 * packs_per_plane;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
     const int cp = (int)(i % packs_per_plane);
     long rest = i / packs_per_plane;
     const int plane = (int)(rest % 9);
-    const long m = rest / 9;
+    const long m = m0 + rest / 9;
     const int wpix = (int)(m % W);
     const int hpix = (int)((m / W) % H);
     const int img = (int)(m / ((long)W * H));
@@ -41,13 +41,14 @@ __global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
 #pragma unroll
       for (int j = 0; j < 8; ++j) v.v[j] = __float2bfloat16(0.f);
     }
-    pstore<bf16, 8>(out + m * (9L * C) + (long)plane * C + cp * 8, v);
+    pstore<bf16, 8>(out + (m - m0) * (9L * C) + (long)plane * C + cp * 8, v);
   }
 }
 
 }  // namespace
 
-torch::Tensor im2col3x3(torch::Tensor x) {
+torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
+                        c10::optional<torch::Tensor> out_buf) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous()
               && x.scalar_type() == torch::kBFloat16);
   const int nd = x.dim();
@@ -56,13 +57,20 @@ torch::Tensor im2col3x3(torch::Tensor x) {
   const int H = x.size(nd - 3), W = x.size(nd - 2), C = x.size(nd - 1);
   TORCH_CHECK(C % 8 == 0, "Cin must be a multiple of 8");
   const long M = (long)IMG * H * W;
-  auto out = torch::empty({M, 9L * C}, x.options());
+  if (m1 < 0) m1 = M;
+  TORCH_CHECK(0 <= m0 && m0 < m1 && m1 <= M);
+  torch::Tensor out;
+  if (out_buf.has_value()) {
+    out = out_buf->narrow(0, 0, m1 - m0);
+  } else {
+    out = torch::empty({m1 - m0, 9L * C}, x.options());
+  }
   const int block = 256;
-  const long total = M * 9 * (C / 8);
+  const long total = (m1 - m0) * 9 * (C / 8);
   const int grid = (int)std::min<long>((total + block - 1) / block, 16384);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(im2col3x3_kernel, dim3(grid), dim3(block), 0, stream,
       reinterpret_cast<const bf16*>(x.data_ptr()),
-      reinterpret_cast<bf16*>(out.data_ptr()), IMG, H, W, C);
+      reinterpret_cast<bf16*>(out.data_ptr()), IMG, H, W, C, m0, m1);
   return out;
 }

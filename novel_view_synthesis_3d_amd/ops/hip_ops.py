@@ -179,16 +179,27 @@ def _miopen_conv(x, w, bias):
     return out.reshape(*shape[:-1], w.shape[0]).contiguous()
 
 
+_WGRAD_CHUNK_BYTES = 96 * 1024 * 1024  # keep the im2col slab L3-resident
+
+
 def _wgrad_im2col_gemm(x, w, dy, has_bias):
-    """Conv wgrad as im2col + ONE hipBLASLt GEMM: dW = A^T @ dy
-    (A = im2col(x), M x 9Cin). The long-K reduction runs at the GEMM peak
-    instead of MIOpen's slow wrw kernels; dbias is a column reduce."""
+    """Conv wgrad as CHUNKED im2col + hipBLASLt GEMMs: dW = sum_c A_c^T dy_c.
+    Chunks are sized so the im2col slab stays inside the 256 MiB Infinity
+    Cache — the GEMM re-reads it from L3, not HBM. Beats MIOpen's wrw
+    kernels on gfx950 (profiles/)."""
     cout, _, _, cin = w.shape
-    A = _OPS.im2col3x3(x)                       # (M, 9Cin) bf16
-    dyf = dy.reshape(-1, cout)                  # (M, Cout)
-    dwf = torch.matmul(A.transpose(0, 1), dyf)  # (9Cin, Cout), fp32 accum
-    # (3,3,Cin,Cout) -> OHWI (Cout,3,3,Cin)
-    dw = dwf.reshape(3, 3, cin, cout).permute(3, 0, 1, 2).contiguous()
+    M = dy.numel() // cout
+    dyf = dy.reshape(M, cout)
+    rows = max(4096, _WGRAD_CHUNK_BYTES // (18 * cin))  # bf16, 9*Cin cols
+    rows = min(rows, M)
+    buf = torch.empty(rows, 9 * cin, device=x.device, dtype=x.dtype)
+    dw32 = None
+    for m0 in range(0, M, rows):
+        m1 = min(M, m0 + rows)
+        A = _OPS.im2col3x3(x, m0, m1, buf)
+        part = torch.matmul(A.transpose(0, 1), dyf[m0:m1])
+        dw32 = part.float() if dw32 is None else dw32.add_(part.float())
+    dw = dw32.to(x.dtype).reshape(3, 3, cin, cout).permute(3, 0, 1, 2)         .contiguous()
     db = dyf.sum(dim=0, dtype=torch.float32).to(dy.dtype) if has_bias else None
     return dw, db
 
