@@ -22,6 +22,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
 torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
                         c10::optional<torch::Tensor> out_buf);
+torch::Tensor attn_p_from_lse(torch::Tensor s, torch::Tensor lse,
+                              double scale);
+torch::Tensor attn_ds(torch::Tensor p, torch::Tensor dp, torch::Tensor delta,
+                      double scale);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -44,6 +48,8 @@ TORCH_LIBRARY(nvs3d, m) {
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
   m.def("im2col3x3(Tensor x, int m0, int m1, Tensor? out_buf) -> Tensor");
+  m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");
+  m.def("attn_ds(Tensor p, Tensor dp, Tensor delta, float scale) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
@@ -54,4 +60,6 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("conv3x3_fwd", conv3x3_fwd);
   m.impl("attn_fwd", attn_fwd);
   m.impl("im2col3x3", im2col3x3);
+  m.impl("attn_p_from_lse", attn_p_from_lse);
+  m.impl("attn_ds", attn_ds);
 }

@@ -240,24 +240,23 @@ class _Attention(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         B, L, H, D = q.shape
         scale = 1.0 / (D ** 0.5)
-        # (B,H,L,D) views
+        # (B,H,L,D) strided views; rocBLAS consumes them without copies
         qt = q.permute(0, 2, 1, 3)
         kt = k.permute(0, 2, 1, 3)
         vt = v.permute(0, 2, 1, 3)
-        ot = o.permute(0, 2, 1, 3).float()
-        dot = do.permute(0, 2, 1, 3).float()
-        s = torch.matmul(qt, kt.transpose(-1, -2)).float() * scale
-        p = torch.exp(s - lse.permute(0, 2, 1)[..., None])  # (B,H,L,Lk)
-        pb = p.to(q.dtype)
-        dv = torch.matmul(pb.transpose(-1, -2), dot.to(q.dtype))
-        dp = torch.matmul(dot.to(q.dtype), vt.transpose(-1, -2)).float()
-        delta = (dot * ot).sum(-1, keepdim=True)
-        ds = (p * (dp - delta)).to(q.dtype)
-        dq = torch.matmul(ds, kt) * scale
-        dk = torch.matmul(ds.transpose(-1, -2), qt) * scale
-        return (dq.permute(0, 2, 1, 3).to(q.dtype),
-                dk.permute(0, 2, 1, 3).to(q.dtype),
-                dv.permute(0, 2, 1, 3).to(q.dtype))
+        dob = do.to(q.dtype).permute(0, 2, 1, 3)
+        # recompute P in ONE pass (bf16 GEMM + fused exp(S*scale - lse))
+        s = torch.matmul(qt, kt.transpose(-1, -2)).contiguous()
+        p = _OPS.attn_p_from_lse(s, lse, scale)        # (B,H,L,Lk) bf16
+        dv = torch.matmul(p.transpose(-1, -2), dob)
+        dp = torch.matmul(dob, vt.transpose(-1, -2)).contiguous()
+        delta = (do.float() * o.float()).sum(-1).permute(0, 2, 1)  # (B,H,L)
+        ds = _OPS.attn_ds(p, dp, delta.contiguous(), scale)
+        dq = torch.matmul(ds, kt)
+        dk = torch.matmul(ds.transpose(-1, -2), qt)
+        return (dq.permute(0, 2, 1, 3).contiguous(),
+                dk.permute(0, 2, 1, 3).contiguous(),
+                dv.permute(0, 2, 1, 3).contiguous())
 
 
 def attention(q, k, v):
