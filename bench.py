@@ -100,6 +100,7 @@ def main() -> None:
                 "parallelism": f"dp{n_gpus}",
                 "final_loss": float(loss.item()),
                 "params": trainer.model.num_params(),
+                "graph": getattr(trainer, "graph_active", False),
                 "peak_mem_gb": round(
                     torch.cuda.max_memory_allocated() / 1e9, 2)
                 if use_cuda else None,

@@ -17,16 +17,37 @@ import numpy as np
 import torch
 
 
+_UP_CACHE = {}
+
+
 def _look_at(eye: torch.Tensor) -> torch.Tensor:
     """cam->world rotation for a camera at `eye` looking at the origin.
     Camera convention: +z forward, +x right, +y down (pinhole with K as in
     data/io.py parse_intrinsics)."""
     fwd = -eye / eye.norm(dim=-1, keepdim=True).clamp_min(1e-8)
-    up = torch.tensor([0.0, 0.0, 1.0], device=eye.device).expand_as(fwd)
+    key = str(eye.device)
+    if key not in _UP_CACHE:
+        _UP_CACHE[key] = torch.tensor([0.0, 0.0, 1.0], device=eye.device)
+    up = _UP_CACHE[key].expand_as(fwd)
     right = torch.cross(fwd, up, dim=-1)
     right = right / right.norm(dim=-1, keepdim=True).clamp_min(1e-8)
     down = torch.cross(fwd, right, dim=-1)
     return torch.stack([right, down, fwd], dim=-1)  # columns = cam axes
+
+
+_K_CACHE = {}
+
+
+def _intrinsics(H: int, device) -> torch.Tensor:
+    """Cached per-(device,H) so the bench's in-graph batch generation stays
+    hipGraph-capturable (no H2D copies inside capture)."""
+    key = (str(device), H)
+    if key not in _K_CACHE:
+        f = 1.75 * H
+        _K_CACHE[key] = torch.tensor(
+            [[f, 0.0, H / 2], [0.0, f, H / 2], [0.0, 0.0, 1.0]],
+            device=device)
+    return _K_CACHE[key]
 
 
 def random_cameras(B: int, H: int, device="cpu",
@@ -41,9 +62,7 @@ def random_cameras(B: int, H: int, device="cpu",
                        r * torch.sin(phi) * torch.sin(theta),
                        r * torch.cos(phi)], dim=-1)
     R = _look_at(eye)
-    f = 1.75 * H
-    K = torch.tensor([[f, 0.0, H / 2], [0.0, f, H / 2], [0.0, 0.0, 1.0]],
-                     device=device).expand(B, 3, 3).contiguous()
+    K = _intrinsics(H, device).expand(B, 3, 3).contiguous()
     return R, eye, K
 
 

@@ -241,15 +241,19 @@ class Trainer:
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph):
             self._static_loss = self._graph_body()
+        self.graph_active = True
+        if self.is_main:
+            print("[trainer] hipGraph captured for the training step",
+                  flush=True)
 
     def _train_step_graphed(self, raw):
         if not hasattr(self, "_graph"):
             try:
                 self._init_graph()
             except Exception as e:
-                import warnings
-                warnings.warn(f"hipGraph capture failed ({e}); falling back "
-                              f"to eager stepping")
+                print(f"[trainer] hipGraph capture FAILED ({e}); falling "
+                      f"back to eager stepping", flush=True)
+                self.graph_active = False
                 self.cfg.use_graph = False
                 torch.cuda.synchronize()
                 return self.train_step(raw)
