@@ -217,9 +217,10 @@ class Trainer:
                  "R2": raw["R2"], "t2": raw["t2"], "K": raw["K"]}
         cond_mask = (torch.rand(B, device=self.device)
                      > self.cfg.cond_drop_prob).to(torch.float32)
-        for p in self.model.parameters():  # zero stable grad storage
-            if p.grad is not None:
-                p.grad.zero_()
+        grads = [p.grad for p in self.model.parameters()
+                 if p.grad is not None]
+        if grads:  # zero the stable grad storage in one fused launch
+            torch._foreach_zero_(grads)
         with self._autocast():
             out = self.model(batch, cond_mask)
         loss = self.compute_loss(out, noise)
