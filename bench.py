@@ -33,7 +33,7 @@ def main() -> None:
     ap.add_argument("--model", default="full")
     ap.add_argument("--sidelength", type=int, default=128)
     ap.add_argument("--amp", default="bf16", choices=["bf16", "off"])
-    ap.add_argument("--graph", default="on", choices=["on", "off"],
+    ap.add_argument("--graph", default="off", choices=["on", "off"],
                     help="hipGraph-capture the train step (single-node)")
     args = ap.parse_args()
 
