@@ -288,6 +288,24 @@ class Trainer:
                 off += n
 
     def train(self) -> None:
+        """Run to train_num_steps. Interrupts (SIGINT/SIGTERM) and crashes
+        checkpoint before exiting, so `--resume auto` continues where the
+        run stopped (the reference loses up to save_every steps on any
+        failure — SURVEY.md §5.3/§5.4)."""
+        try:
+            self._train_loop()
+        except (KeyboardInterrupt, Exception) as e:
+            if self.is_main and self.step > 0:
+                path = ckpt.save_checkpoint(
+                    self.cfg.ckpt_folder, self.model, self.opt, self.step,
+                    extra={"model_cfg": vars(self.model_cfg),
+                           "img_sidelength": self.cfg.img_sidelength,
+                           "interrupted": repr(e)})
+                print(f"[trainer] interrupted at step {self.step}; "
+                      f"checkpoint saved to {path}", flush=True)
+            raise
+
+    def _train_loop(self) -> None:
         cfg = self.cfg
         while self.step < cfg.train_num_steps:
             t0 = time.perf_counter()

@@ -37,6 +37,8 @@ def main() -> None:
     ap.add_argument("--resume", default=None,
                     help="checkpoint path, or 'auto'")
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--profile", action="store_true",
+                    help="torch.profiler trace of 5 steps, then exit")
     args = ap.parse_args()
 
     if args.config:
@@ -60,6 +62,12 @@ def main() -> None:
                       results_folder=args.results_folder,
                       model_cfg=model_cfg,
                       train_cfg=train_cfg)
+    if args.profile:
+        from novel_view_synthesis_3d_amd.engine.profiling import (
+            profile_training,
+        )
+        profile_training(trainer)
+        return
     trainer.train()
 
 
