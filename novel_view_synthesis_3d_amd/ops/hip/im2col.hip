@@ -16,9 +16,12 @@ namespace {
 
 using bf16 = __hip_bfloat16;
 
+// m indexes the OUTPUT grid (IMG, Ho, Wo); input pixel = out*stride + d - pad
 __global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
                                  bf16* __restrict__ out,      // (m1-m0, 9*C)
                                  int IMG, int H, int W, int C,
+                                 int Ho, int Wo, int stride_s,
+                                 int pad_h, int pad_w,
                                  long m0, long m1) {
   const int packs_per_plane = C / 8;
   const long total = (m1 - m0) * 9 * packs_per_plane;
@@ -29,11 +32,11 @@ __global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
     long rest = i / packs_per_plane;
     const int plane = (int)(rest % 9);
     const long m = m0 + rest / 9;
-    const int wpix = (int)(m % W);
-    const int hpix = (int)((m / W) % H);
-    const int img = (int)(m / ((long)W * H));
-    const int hh = hpix + plane / 3 - 1;
-    const int ww = wpix + plane % 3 - 1;
+    const int wpix = (int)(m % Wo);
+    const int hpix = (int)((m / Wo) % Ho);
+    const int img = (int)(m / ((long)Wo * Ho));
+    const int hh = hpix * stride_s + plane / 3 - pad_h;
+    const int ww = wpix * stride_s + plane % 3 - pad_w;
     Pack<bf16, 8> v;
     if (hh >= 0 && hh < H && ww >= 0 && ww < W) {
       v = pload<bf16, 8>(x + (((long)img * H + hh) * W + ww) * C + cp * 8);
@@ -47,7 +50,8 @@ __global__ void im2col3x3_kernel(const bf16* __restrict__ x,  // (IMG,H,W,C)
 
 }  // namespace
 
-torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
+torch::Tensor im2col3x3(torch::Tensor x, int64_t stride_s,
+                        int64_t m0, int64_t m1,
                         c10::optional<torch::Tensor> out_buf) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous()
               && x.scalar_type() == torch::kBFloat16);
@@ -56,7 +60,12 @@ torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
   const int IMG = nd == 5 ? x.size(0) * x.size(1) : x.size(0);
   const int H = x.size(nd - 3), W = x.size(nd - 2), C = x.size(nd - 1);
   TORCH_CHECK(C % 8 == 0, "Cin must be a multiple of 8");
-  const long M = (long)IMG * H * W;
+  const int s = (int)stride_s;
+  // FLAX SAME: out = ceil(H/s), pad_lo = total//2 (asymmetric)
+  const int Ho = (H + s - 1) / s, Wo = (W + s - 1) / s;
+  const int pad_h = std::max((Ho - 1) * s + 3 - H, 0) / 2;
+  const int pad_w = std::max((Wo - 1) * s + 3 - W, 0) / 2;
+  const long M = (long)IMG * Ho * Wo;
   if (m1 < 0) m1 = M;
   TORCH_CHECK(0 <= m0 && m0 < m1 && m1 <= M);
   torch::Tensor out;
@@ -71,6 +80,7 @@ torch::Tensor im2col3x3(torch::Tensor x, int64_t m0, int64_t m1,
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(im2col3x3_kernel, dim3(grid), dim3(block), 0, stream,
       reinterpret_cast<const bf16*>(x.data_ptr()),
-      reinterpret_cast<bf16*>(out.data_ptr()), IMG, H, W, C, m0, m1);
+      reinterpret_cast<bf16*>(out.data_ptr()), IMG, H, W, C,
+      Ho, Wo, s, pad_h, pad_w, m0, m1);
   return out;
 }

@@ -196,7 +196,7 @@ def _wgrad_im2col_gemm(x, w, dy, has_bias):
     dw32 = None
     for m0 in range(0, M, rows):
         m1 = min(M, m0 + rows)
-        A = _OPS.im2col3x3(x, m0, m1, buf)
+        A = _OPS.im2col3x3(x, 1, m0, m1, buf)
         part = torch.matmul(A.transpose(0, 1), dyf[m0:m1])
         dw32 = part.float() if dw32 is None else dw32.add_(part.float())
     dw = dw32.to(x.dtype).reshape(3, 3, cin, cout).permute(3, 0, 1, 2)         .contiguous()
@@ -216,8 +216,85 @@ def _miopen_wgrad(x, w, dy, has_bias):
     return dw_ohwi, db
 
 
+def _pad_c8(t):
+    """Pad the channel (last) dim to a multiple of 8 (im2col needs 16B packs)."""
+    c = t.shape[-1]
+    pad = (-c) % 8
+    if pad == 0:
+        return t, c
+    import torch.nn.functional as Fn
+    return Fn.pad(t, (0, pad)), c
+
+
+def _gemm_conv_fwd(x, w, bias, stride):
+    """Generic 3x3 SAME conv for shapes the MFMA igemm kernel doesn't take
+    (stem 3ch, pose-emb 144ch strided, head 3ch, small configs): strided
+    im2col kernel + ONE hipBLASLt GEMM. Replaces MIOpen entirely."""
+    cout, _, _, cin = w.shape
+    xp, _ = _pad_c8(x.contiguous())
+    A = _OPS.im2col3x3(xp, stride, 0, -1, None)          # (Mo, 9*Cpad)
+    cpad = xp.shape[-1]
+    w2 = w.permute(1, 2, 3, 0)                            # (3,3,Cin,Cout)
+    if cpad != cin:
+        import torch.nn.functional as Fn
+        w2 = Fn.pad(w2, (0, 0, 0, cpad - cin))
+    wk = w2.reshape(9 * cpad, cout)
+    y = torch.matmul(A, wk.to(A.dtype))
+    if bias is not None:
+        y = y + bias.to(y.dtype)
+    H, W = x.shape[-3], x.shape[-2]
+    Ho, Wo = -(-H // stride), -(-W // stride)
+    return y.reshape(*x.shape[:-3], Ho, Wo, cout)
+
+
+class _FrameConvGeneric(torch.autograd.Function):
+    """Fallback conv shapes; fwd/wgrad via im2col+GEMM, dgrad via im2col of
+    dy (stride 1 only — the model's strided convs consume no-grad inputs)."""
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x, weight, bias, stride):
+        x = x.contiguous()
+        weight = weight.contiguous()
+        y = _gemm_conv_fwd(x, weight, bias, stride)
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        ctx.stride = stride
+        return y
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        cout, _, _, cin = w.shape
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            assert ctx.stride == 1, "strided dgrad not needed by the model"
+            wt = torch.flip(w, dims=(1, 2)).permute(3, 1, 2, 0).contiguous()
+            if conv_shapes_supported(cout, cin, 1):
+                dx = _OPS.conv3x3_fwd(dy, wt, None)
+            else:
+                dx = _gemm_conv_fwd(dy, wt, None, 1)
+        if ctx.needs_input_grad[1] or ctx.has_bias:
+            dyf = dy.reshape(-1, cout)
+            xp, _ = _pad_c8(x)
+            cpad = xp.shape[-1]
+            A = _OPS.im2col3x3(xp, ctx.stride, 0, -1, None)
+            dwf = torch.matmul(A.transpose(0, 1), dyf)    # (9*Cpad, Cout)
+            dw = dwf.reshape(3, 3, cpad, cout)[:, :, :cin, :]                 .permute(3, 0, 1, 2).contiguous()
+            if not ctx.needs_input_grad[1]:
+                dw = None
+            db = (dyf.sum(0, dtype=torch.float32).to(dy.dtype)
+                  if ctx.has_bias else None)
+        return dx, dw, db, None
+
+
 def frame_conv3x3(x, weight, bias, stride: int = 1):
-    return _FrameConv3x3.apply(x, weight, bias)
+    cout, _, _, cin = weight.shape
+    if conv_shapes_supported(cin, cout, stride):
+        return _FrameConv3x3.apply(x, weight, bias)
+    return _FrameConvGeneric.apply(x, weight, bias, stride)
 
 
 # ---------------------------------------------------------------------------

@@ -278,3 +278,78 @@ def test_attention_autograd_parity():
         err = (gg - ww).abs().max().item()
         scale = ww.abs().max().item() + 1e-6
         assert err / scale < 6e-2, f"{n}: rel {err/scale:.3e}"
+
+
+@pytest.mark.parametrize("shape,stride", [
+    ((2, 2, 32, 32, 3, 256), 1),    # stem-like: Cin=3 (padded to 8)
+    ((2, 2, 32, 32, 144, 128), 2),  # pose-emb-like: strided SAME
+    ((1, 2, 32, 32, 144, 128), 4),
+    ((1, 2, 32, 32, 256, 3), 1),    # head-like: Cout=3
+])
+def test_generic_gemm_conv_forward_parity(shape, stride):
+    B, F, H, W, Cin, Cout = shape
+    g = torch.Generator(device="cuda").manual_seed(0)
+    x = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                    dtype=torch.bfloat16)
+    w = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                    dtype=torch.bfloat16) * (1.0 / (3 * max(Cin, 8) ** 0.5))
+    b = torch.randn(Cout, device="cuda", generator=g) * 0.1
+    got = hip_ops.frame_conv3x3(x, w, b, stride)
+    want = ref.frame_conv3x3(x.float(), w.float(), b.float(), stride)
+    assert got.shape == want.shape
+    err = (got.float() - want).abs().max().item()
+    scale = want.abs().max().item() + 1e-6
+    assert err / scale < 3e-2, (err, scale)
+
+
+def test_generic_gemm_conv_head_autograd():
+    """Head conv (Cout=3): dgrad via im2col of dy, wgrad via im2col GEMM."""
+    B, F, H, W, Cin, Cout = 1, 2, 16, 16, 128, 3
+    g = torch.Generator(device="cuda").manual_seed(3)
+    x0 = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    w0 = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.05
+    b0 = torch.randn(Cout, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.1
+    dy = torch.randn(B, F, H, W, Cout, device="cuda", generator=g)
+
+    def run(fn, dtype):
+        x = x0.detach().to(dtype).requires_grad_(True)
+        w = w0.detach().to(dtype).requires_grad_(True)
+        b = b0.detach().to(dtype).requires_grad_(True)
+        y = fn(x, w, b, 1)
+        (y.float() * dy).sum().backward()
+        return y.float(), x.grad.float(), w.grad.float(), b.grad.float()
+
+    got = run(hip_ops.frame_conv3x3, torch.bfloat16)
+    want = run(ref.frame_conv3x3, torch.float32)
+    for n, gg, ww in zip(["y", "dx", "dw", "db"], got, want):
+        err = (gg - ww).abs().max().item()
+        scale = ww.abs().max().item() + 1e-6
+        assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
+
+
+def test_strided_conv_wgrad_parity():
+    """Pose-emb convs train their weights through the strided path."""
+    B, F, H, W, Cin, Cout = 1, 2, 16, 16, 144, 64
+    g = torch.Generator(device="cuda").manual_seed(4)
+    x0 = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    w0 = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.05
+    dy = torch.randn(B, F, H // 2, W // 2, Cout, device="cuda", generator=g)
+
+    def run(fn, dtype):
+        x = x0.detach().to(dtype)  # no grad for x (pose path)
+        w = w0.detach().to(dtype).requires_grad_(True)
+        y = fn(x, w, None, 2)
+        (y.float() * dy).sum().backward()
+        return y.float(), w.grad.float()
+
+    got = run(hip_ops.frame_conv3x3, torch.bfloat16)
+    want = run(ref.frame_conv3x3, torch.float32)
+    for n, gg, ww in zip(["y", "dw"], got, want):
+        err = (gg - ww).abs().max().item()
+        scale = ww.abs().max().item() + 1e-6
+        assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
