@@ -294,6 +294,15 @@ def frame_conv3x3(x, weight, bias, stride: int = 1):
     cout, _, _, cin = weight.shape
     if conv_shapes_supported(cin, cout, stride):
         return _FrameConv3x3.apply(x, weight, bias)
+    if stride == 1 and cout % 128 == 0 and cin >= 64:
+        # e.g. the 144ch stride-1 pose conv: zero-pad channels to %64 and
+        # take the MFMA kernel (F.pad is differentiable; the pad columns
+        # multiply zero weights). Far cheaper than a 9x im2col of the input.
+        import torch.nn.functional as Fn
+        pad = (-cin) % 64
+        xp = Fn.pad(x, (0, pad))
+        wp = Fn.pad(weight, (0, pad))
+        return _FrameConv3x3.apply(xp, wp, bias)
     return _FrameConvGeneric.apply(x, weight, bias, stride)
 
 
