@@ -230,21 +230,46 @@ class ResnetBlock(nn.Module):
 class AttnLayer(nn.Module):
     """q/k/v projections + multi-head attention
     (/root/reference/model/xunet.py:94-103). NOTE: no output projection —
-    the reference comments it out (xunet.py:126)."""
+    the reference comments it out (xunet.py:126).
+
+    The three DenseGenerals are fused into ONE projection GEMM at call time
+    (weights concatenated; q/k/v become strided views of the (B,L,3C) output
+    — the MFMA attention kernel consumes the views directly)."""
 
     flax_type = "AttnLayer"
 
     def __init__(self, channels: int, heads: int):
         super().__init__()
-        head_dim = channels // heads
+        self.heads = heads
+        self.head_dim = channels // heads
+        head_dim = self.head_dim
         self.DenseGeneral_0 = DenseGeneral(channels, heads, head_dim)  # q
         self.DenseGeneral_1 = DenseGeneral(channels, heads, head_dim)  # k
         self.DenseGeneral_2 = DenseGeneral(channels, heads, head_dim)  # v
 
+    def project_qkv(self, x: torch.Tensor):
+        """(B,L,C) -> q,k,v (B,L,h,d) strided views of one fused GEMM."""
+        C = x.shape[-1]
+        w = torch.cat([self.DenseGeneral_0.weight,
+                       self.DenseGeneral_1.weight,
+                       self.DenseGeneral_2.weight], dim=0)
+        b = torch.cat([self.DenseGeneral_0.bias, self.DenseGeneral_1.bias,
+                       self.DenseGeneral_2.bias], dim=0)
+        y = F.linear(x, w, b)  # (B, L, 3C)
+        hd = (self.heads, self.head_dim)
+        return (y[..., :C].unflatten(-1, hd),
+                y[..., C:2 * C].unflatten(-1, hd),
+                y[..., 2 * C:].unflatten(-1, hd))
+
     def forward(self, q: torch.Tensor, kv: torch.Tensor) -> torch.Tensor:
         B, L, C = q.shape
-        out = ops.attention(self.DenseGeneral_0(q), self.DenseGeneral_1(kv),
-                            self.DenseGeneral_2(kv))
+        if q is kv:
+            qq, kk, vv = self.project_qkv(q)
+        else:
+            qq = self.DenseGeneral_0(q)
+            kk = self.DenseGeneral_1(kv)
+            vv = self.DenseGeneral_2(kv)
+        out = ops.attention(qq, kk, vv)
         return out.reshape(B, L, C)
 
 
@@ -268,13 +293,17 @@ class AttnBlock(nn.Module):
         h = self.GroupNorm_0(h_in)
         h0 = h[:, 0].reshape(B, H * W, C)
         h1 = h[:, 1].reshape(B, H * W, C)
+        # ONE fused QKV projection per frame (the reference projects q and
+        # k/v separately per call — 6 GEMMs where 2 suffice); cross-frame
+        # attention reuses the other frame's k/v views.
+        q0, k0, v0 = self.AttnLayer_0.project_qkv(h0)
+        q1, k1, v1 = self.AttnLayer_0.project_qkv(h1)
         if self.attn_type == "self":
-            h0 = self.AttnLayer_0(q=h0, kv=h0)
-            h1 = self.AttnLayer_0(q=h1, kv=h1)
+            h0 = ops.attention(q0, k0, v0).reshape(B, H * W, C)
+            h1 = ops.attention(q1, k1, v1).reshape(B, H * W, C)
         else:
-            h0_orig = h0
-            h0 = self.AttnLayer_0(q=h0, kv=h1)
-            h1 = self.AttnLayer_0(q=h1, kv=h0_orig)
+            h0 = ops.attention(q0, k1, v1).reshape(B, H * W, C)
+            h1 = ops.attention(q1, k0, v0).reshape(B, H * W, C)
         h = torch.stack([h0, h1], dim=1).reshape(B, Fr, H, W, C)
         return ops.residual_scale_add(h, h_in)
 

@@ -36,9 +36,14 @@ constexpr int WAVES = 4;   // q rows per block = 64
 constexpr float NEG_INF = -1e30f;
 
 struct AttnShape {
-  int B, L, H, D;       // (B, L, H, D) layout, token stride = H*D
+  int B, L, H, D;
   int Lk;               // key sequence length (same L here)
   float scale;
+  // strides in ELEMENTS (d contiguous, head stride = D): lets q/k/v be
+  // views into one fused QKV projection (B, L, 3C) without copies
+  long sqb, sqt;        // q batch, token
+  long skb, skt;        // k
+  long svb, svt;        // v
 };
 
 __device__ __forceinline__ int swz_row(int row, int byte_in_row,
@@ -73,10 +78,10 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int head = bid % s.H;
   const int b = bid / s.H;
 
-  const int HD = s.H * s.D;
-  const bf16* qbase = q + ((long)b * s.L) * HD + head * s.D;
-  const bf16* kbase = k + ((long)b * s.Lk) * HD + head * s.D;
-  const bf16* vbase = v + ((long)b * s.Lk) * HD + head * s.D;
+  const int HD = s.H * s.D;  // output layout stays contiguous (B,L,H,D)
+  const bf16* qbase = q + (long)b * s.sqb + head * s.D;
+  const bf16* kbase = k + (long)b * s.skb + head * s.D;
+  const bf16* vbase = v + (long)b * s.svb + head * s.D;
 
   // ---- load Q into A-fragment registers, pre-scaled ----
   // chunk dc: lane l holds Q[q0 + (l&15)][dc*32 + (l>>4)*8 .. +8]
@@ -87,7 +92,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
   for (int dc = 0; dc < DC; ++dc) {
     const int off = dc * 32 + (lane >> 4) * 8;
     if (off < s.D) {
-      Pack<bf16, 8> p = pload<bf16, 8>(qbase + (long)qrow * HD + off);
+      Pack<bf16, 8> p = pload<bf16, 8>(qbase + (long)qrow * s.sqt + off);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         qf[dc][j] = (__bf16)__float2bfloat16(
@@ -122,12 +127,12 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
           const int key = p / (D / 8);
           const int d0 = (p % (D / 8)) * 8;
           Pack<bf16, 8> kv8 =
-              pload<bf16, 8>(kbase + (long)(t * KB + key) * HD + d0);
+              pload<bf16, 8>(kbase + (long)(t * KB + key) * s.skt + d0);
           *reinterpret_cast<Pack<bf16, 8>*>(
               ldsK + swz_row(key, d0 * 2, D * 2)) = kv8;
           // V: same source geometry, transposed scatter into [D][KB]
           Pack<bf16, 8> vv8 =
-              pload<bf16, 8>(vbase + (long)(t * KB + key) * HD + d0);
+              pload<bf16, 8>(vbase + (long)(t * KB + key) * s.svt + d0);
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             *reinterpret_cast<bf16*>(
@@ -268,10 +273,15 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v) {
-  // q/k/v: (B, L, H, D) bf16 contiguous
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous()
-              && v.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  // q/k/v: (B, L, H, D) bf16 — may be strided VIEWS (e.g. into a fused
+  // (B, L, 3C) QKV projection) as long as d is contiguous and the head
+  // stride is D.
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  auto check_strides = [](const torch::Tensor& t) {
+    TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == t.size(3),
+                "attn operand needs contiguous (head, d) tail");
+  };
+  check_strides(q); check_strides(k); check_strides(v);
   const int B = q.size(0), L = q.size(1), H = q.size(2), D = q.size(3);
   TORCH_CHECK(k.size(1) == L, "self/cross attention has equal q/kv length");
   TORCH_CHECK(L % 64 == 0, "L must be a multiple of 64");
@@ -281,6 +291,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   AttnShape s;
   s.B = B; s.L = L; s.H = H; s.D = D; s.Lk = k.size(1);
   s.scale = 1.0f / std::sqrt((float)D);
+  s.sqb = q.stride(0); s.sqt = q.stride(1);
+  s.skb = k.stride(0); s.skt = k.stride(1);
+  s.svb = v.stride(0); s.svt = v.stride(1);
 
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, L, H}, q.options().dtype(torch::kFloat));

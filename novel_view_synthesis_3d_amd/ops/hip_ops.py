@@ -315,7 +315,14 @@ class _Attention(torch.autograd.Function):
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
     def forward(ctx, q, k, v):
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # strided views (fused-QKV slices) are consumed directly; the
+        # kernel requires only a contiguous (head, d) tail
+        if q.stride(-1) != 1 or q.stride(2) != q.size(3):
+            q = q.contiguous()
+        if k.stride(-1) != 1 or k.stride(2) != k.size(3):
+            k = k.contiguous()
+        if v.stride(-1) != 1 or v.stride(2) != v.size(3):
+            v = v.contiguous()
         out, lse = _OPS.attn_fwd(q, k, v)
         ctx.save_for_backward(q, k, v, out, lse)
         return out

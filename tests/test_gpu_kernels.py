@@ -353,3 +353,23 @@ def test_strided_conv_wgrad_parity():
         err = (gg - ww).abs().max().item()
         scale = ww.abs().max().item() + 1e-6
         assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
+
+
+def test_attention_strided_qkv_views():
+    """The kernel consumes q/k/v as views into one fused (B,L,3C) GEMM
+    output without copies — must equal the contiguous path."""
+    B, L, h, d = 2, 256, 4, 128
+    C = h * d
+    g = torch.Generator(device="cuda").manual_seed(5)
+    qkv = torch.randn(B, L, 3 * C, device="cuda", generator=g,
+                      dtype=torch.bfloat16)
+    q = qkv[..., :C].unflatten(-1, (h, d))
+    k = qkv[..., C:2 * C].unflatten(-1, (h, d))
+    v = qkv[..., 2 * C:].unflatten(-1, (h, d))
+    out_s, lse_s = torch.ops.nvs3d.attn_fwd(q, k, v)
+    out_c, lse_c = torch.ops.nvs3d.attn_fwd(q.contiguous(), k.contiguous(),
+                                            v.contiguous())
+    assert torch.equal(out_s, out_c)
+    assert torch.equal(lse_s, lse_c)
+    want = ref.attention(q.float(), k.float(), v.float())
+    assert (out_s.float() - want).abs().max().item() < 2e-2
