@@ -221,10 +221,13 @@ class ResnetBlock(nn.Module):
         h = self.GroupNorm_1(h, film=film, silu=True)
         if self.dropout_rate > 0.0:
             h = F.dropout(h, self.dropout_rate, training=self.training)
-        h = self.Conv_1(h)
         if self.Dense_0 is not None:
             h_in = self.Dense_0(h_in)
-        return ops.residual_scale_add(h, h_in)
+        # residual + 1/sqrt(2) fused into the conv epilogue (xunet.py:92)
+        return ops.frame_conv3x3_residual(
+            h, self.Conv_1.weight, self.Conv_1.bias,
+            h_in.to(h.dtype) if h_in.dtype != h.dtype else h_in,
+            ops.SQRT_HALF)
 
 
 class AttnLayer(nn.Module):

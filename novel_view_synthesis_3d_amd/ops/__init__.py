@@ -116,6 +116,18 @@ def residual_scale_add(h, h_in):
     return ref.residual_scale_add(h, h_in)
 
 
+def frame_conv3x3_residual(x, weight, bias, residual, res_scale: float):
+    """Fused ResnetBlock tail: (conv(x,w)+bias+residual)*res_scale."""
+    if _use_hip(x, "frame_conv3x3"):
+        bf16_path = (x.dtype == torch.bfloat16
+                     or torch.is_autocast_enabled())
+        if bf16_path:
+            return _HIP_MOD.frame_conv3x3_residual(x, weight, bias,
+                                                   residual, res_scale)
+    y = ref.frame_conv3x3(x, weight, bias, 1)
+    return (y + residual) * res_scale
+
+
 def pose_embedding(R, t, K, cond_mask, H: int, W: int, out_dtype):
     if _use_hip(R, "pose_embedding"):
         return _HIP_MOD.pose_embedding(R, t, K, cond_mask, H, W, out_dtype)

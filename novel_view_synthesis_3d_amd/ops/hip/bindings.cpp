@@ -17,7 +17,9 @@ void fused_adam(torch::Tensor ptrs, torch::Tensor chunk_tensor,
                 torch::Tensor chunk_off, torch::Tensor numels,
                 double lr, double b1, double b2, double eps, int64_t step);
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
-                          c10::optional<torch::Tensor> bias);
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> residual,
+                          double out_scale);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
 torch::Tensor im2col3x3(torch::Tensor x, int64_t stride, int64_t m0,
@@ -45,7 +47,7 @@ TORCH_LIBRARY(nvs3d, m) {
   m.def("fused_adam(Tensor ptrs, Tensor chunk_tensor, Tensor chunk_off, "
         "Tensor numels, float lr, float b1, float b2, float eps, "
         "int step) -> ()");
-  m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias) -> Tensor");
+  m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias, Tensor? residual, float out_scale) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
   m.def("im2col3x3(Tensor x, int stride, int m0, int m1, Tensor? out_buf) -> Tensor");
   m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");

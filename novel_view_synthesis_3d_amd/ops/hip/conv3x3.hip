@@ -65,6 +65,8 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
                    const bf16* __restrict__ w,   // (Cout, 9*Cin) rows=co
                    const float* __restrict__ bias,  // (Cout,) or null
                    const bf16* __restrict__ zbuf,   // 128B of zeros (padding)
+                   const bf16* __restrict__ residual,  // out-shaped or null
+                   float out_scale,              // y=(conv+bias+res)*scale
                    bf16* __restrict__ out,       // (IMG,H,W,Cout)
                    ConvShape s, int nblocks_m) {
   // XCD-aware block swizzle (bijective form, §5.5 T1): consecutive
@@ -211,6 +213,8 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
   }
   __syncthreads();
   // store: 128 rows x 16 packs(16B) = 2048 packs / 256 threads = 8 each
+  // optional fused residual: y = (conv + bias + residual) * out_scale
+  // (reference ResnetBlock tail, xunet.py:92)
 #pragma unroll
   for (int it = 0; it < 8; ++it) {
     const int p = tid + it * THREADS;
@@ -220,7 +224,20 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
     if (m < s.M) {
       Pack<bf16, 8> v = *reinterpret_cast<Pack<bf16, 8>*>(
           ldsC + row * BN + cp * 8);
-      pstore<bf16, 8>(out + (long)m * s.Cout + n0 + cp * 8, v);
+      const long goff = (long)m * s.Cout + n0 + cp * 8;
+      if (residual != nullptr) {
+        Pack<bf16, 8> rv = pload<bf16, 8>(residual + goff);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          from_f32((to_f32(v.v[j]) + to_f32(rv.v[j])) * out_scale, v.v[j]);
+        }
+      } else if (out_scale != 1.0f) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          from_f32(to_f32(v.v[j]) * out_scale, v.v[j]);
+        }
+      }
+      pstore<bf16, 8>(out + goff, v);
     }
   }
 }
@@ -228,7 +245,9 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
 }  // namespace
 
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
-                          c10::optional<torch::Tensor> bias) {
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> residual,
+                          double out_scale) {
   // x: (B,F,H,W,Cin) or (IMG,H,W,Cin) bf16 contiguous; w: (Cout,3,3,Cin)
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
@@ -262,11 +281,19 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
   const int grid = nblocks_m * (s.Cout / BN);
   const size_t lds = 2 * (BM + BN) * BK * 2;  // 64 KB
   auto stream = at::hip::getCurrentHIPStream();
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous()
+                && residual->sizes() == out.sizes()
+                && residual->scalar_type() == torch::kBFloat16);
+  }
   hipLaunchKernelGGL(conv3x3_igemm, dim3(grid), dim3(THREADS), lds, stream,
       reinterpret_cast<const bf16*>(x.data_ptr()),
       reinterpret_cast<const bf16*>(w.data_ptr()),
       bias.has_value() ? biasf.data_ptr<float>() : nullptr,
       reinterpret_cast<const bf16*>(zbuf.data_ptr()),
+      residual.has_value()
+          ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr,
+      (float)out_scale,
       reinterpret_cast<bf16*>(out.data_ptr()), s, nblocks_m);
   return out;
 }

@@ -130,14 +130,22 @@ def fused_adam_step(plan: AdamPlan, lr, b1, b2, eps, step: int) -> None:
 # ---------------------------------------------------------------------------
 
 class _FrameConv3x3(torch.autograd.Function):
+    """MFMA igemm conv with an optionally fused residual tail:
+    y = (conv(x, w) + bias + residual) * res_scale (the ResnetBlock ending,
+    reference xunet.py:92)."""
+
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
-    def forward(ctx, x, weight, bias):
+    def forward(ctx, x, weight, bias, residual, res_scale):
         x = x.contiguous()
         weight = weight.contiguous()
-        y = _OPS.conv3x3_fwd(x, weight, bias)
+        if residual is not None:
+            residual = residual.contiguous()
+        y = _OPS.conv3x3_fwd(x, weight, bias, residual, res_scale)
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
+        ctx.has_res = residual is not None
+        ctx.res_scale = res_scale
         return y
 
     @staticmethod
@@ -145,6 +153,8 @@ class _FrameConv3x3(torch.autograd.Function):
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
         dy = dy.contiguous()
+        if ctx.res_scale != 1.0:
+            dy = dy * ctx.res_scale
         cout, _, _, cin = w.shape
         need_x, need_w, need_b = (ctx.needs_input_grad[0],
                                   ctx.needs_input_grad[1], ctx.has_bias)
@@ -153,14 +163,15 @@ class _FrameConv3x3(torch.autograd.Function):
             # dgrad = conv3x3(dy, w~), w~[ci,ey,ex,co] = w[co,2-ey,2-ex,ci]
             wt = torch.flip(w, dims=(1, 2)).permute(3, 1, 2, 0).contiguous()
             if conv_shapes_supported(cout, cin, 1):
-                dx = _OPS.conv3x3_fwd(dy, wt, None)
+                dx = _OPS.conv3x3_fwd(dy, wt, None, None, 1.0)
             else:
                 dx = _miopen_conv(dy, wt, None)
         if need_w or need_b:
             dw4, db = _wgrad_im2col_gemm(x, w, dy, ctx.has_bias)
             if need_w:
                 dw = dw4
-        return dx, dw, db
+        dres = dy if ctx.has_res else None
+        return dx, dw, db, dres, None
 
 
 def _nchw_view(x5):
@@ -273,7 +284,7 @@ class _FrameConvGeneric(torch.autograd.Function):
             assert ctx.stride == 1, "strided dgrad not needed by the model"
             wt = torch.flip(w, dims=(1, 2)).permute(3, 1, 2, 0).contiguous()
             if conv_shapes_supported(cout, cin, 1):
-                dx = _OPS.conv3x3_fwd(dy, wt, None)
+                dx = _OPS.conv3x3_fwd(dy, wt, None, None, 1.0)
             else:
                 dx = _gemm_conv_fwd(dy, wt, None, 1)
         if ctx.needs_input_grad[1] or ctx.has_bias:
@@ -293,7 +304,7 @@ class _FrameConvGeneric(torch.autograd.Function):
 def frame_conv3x3(x, weight, bias, stride: int = 1):
     cout, _, _, cin = weight.shape
     if conv_shapes_supported(cin, cout, stride):
-        return _FrameConv3x3.apply(x, weight, bias)
+        return _FrameConv3x3.apply(x, weight, bias, None, 1.0)
     if stride == 1 and cout % 128 == 0 and cin >= 64:
         # e.g. the 144ch stride-1 pose conv: zero-pad channels to %64 and
         # take the MFMA kernel (F.pad is differentiable; the pad columns
@@ -302,7 +313,7 @@ def frame_conv3x3(x, weight, bias, stride: int = 1):
         pad = (-cin) % 64
         xp = Fn.pad(x, (0, pad))
         wp = Fn.pad(weight, (0, pad))
-        return _FrameConv3x3.apply(xp, wp, bias)
+        return _FrameConv3x3.apply(xp, wp, bias, None, 1.0)
     return _FrameConvGeneric.apply(x, weight, bias, stride)
 
 
@@ -354,3 +365,13 @@ class _Attention(torch.autograd.Function):
 
 def attention(q, k, v):
     return _Attention.apply(q, k, v)
+
+
+def frame_conv3x3_residual(x, weight, bias, residual, res_scale):
+    """y = (conv3x3(x,w) + bias + residual) * res_scale, residual fused in
+    the conv epilogue when the MFMA kernel covers the shape."""
+    cout, _, _, cin = weight.shape
+    if conv_shapes_supported(cin, cout, 1):
+        return _FrameConv3x3.apply(x, weight, bias, residual, res_scale)
+    y = frame_conv3x3(x, weight, bias, 1)
+    return (y + residual) * res_scale

@@ -373,3 +373,40 @@ def test_attention_strided_qkv_views():
     assert torch.equal(lse_s, lse_c)
     want = ref.attention(q.float(), k.float(), v.float())
     assert (out_s.float() - want).abs().max().item() < 2e-2
+
+
+def test_conv_residual_fused_tail():
+    """y = (conv(x,w)+b+res)/sqrt(2) fused in the epilogue — fwd + grads."""
+    B, F, H, W, Cin, Cout = 1, 2, 16, 16, 256, 128
+    g = torch.Generator(device="cuda").manual_seed(6)
+    x0 = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    w0 = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.02
+    b0 = torch.randn(Cout, device="cuda", generator=g,
+                     dtype=torch.bfloat16) * 0.1
+    r0 = torch.randn(B, F, H, W, Cout, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    dy = torch.randn(B, F, H, W, Cout, device="cuda", generator=g)
+    import math
+    sc = 1.0 / math.sqrt(2.0)
+
+    def run(fused, dtype):
+        x = x0.detach().to(dtype).requires_grad_(True)
+        w = w0.detach().to(dtype).requires_grad_(True)
+        b = b0.detach().to(dtype).requires_grad_(True)
+        r = r0.detach().to(dtype).requires_grad_(True)
+        if fused:
+            y = hip_ops.frame_conv3x3_residual(x, w, b, r, sc)
+        else:
+            y = (ref.frame_conv3x3(x, w, b) + r) * sc
+        (y.float() * dy).sum().backward()
+        return y.float(), x.grad.float(), w.grad.float(), b.grad.float(), \
+            r.grad.float()
+
+    got = run(True, torch.bfloat16)
+    want = run(False, torch.float32)
+    for n, gg, ww in zip(["y", "dx", "dw", "db", "dres"], got, want):
+        err = (gg - ww).abs().max().item()
+        scale = ww.abs().max().item() + 1e-6
+        assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
