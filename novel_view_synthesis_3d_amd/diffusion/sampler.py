@@ -18,7 +18,6 @@ Redesign:
 
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional
 
 import torch

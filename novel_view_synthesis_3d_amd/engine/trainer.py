@@ -15,7 +15,6 @@ Keeps the reference's public surface (Trainer(folder, train_batch_size=...,
 from __future__ import annotations
 
 import contextlib
-import math
 import os
 import time
 from pathlib import Path

@@ -19,7 +19,7 @@ import torch.nn.functional as F
 from torch import nn
 
 from novel_view_synthesis_3d_amd import ops
-from novel_view_synthesis_3d_amd.utils.init import lecun_normal_, variance_scaling_
+from novel_view_synthesis_3d_amd.utils.init import lecun_normal_
 
 
 # ---------------------------------------------------------------------------

@@ -34,7 +34,6 @@ from novel_view_synthesis_3d_amd.config import XUNetConfig
 from novel_view_synthesis_3d_amd.models.layers import (
     AttnBlock, Dense, FrameConv, JointGroupNorm, ResnetBlock, XUNetBlock,
 )
-from novel_view_synthesis_3d_amd.models.rays import camera_rays
 from novel_view_synthesis_3d_amd.utils.init import normal_
 
 POSE_EMB_DIM = 3 * (1 + 2 * 15) + 3 * (1 + 2 * 8)  # 93 + 51 = 144
