@@ -179,7 +179,7 @@ def test_conv3x3_forward_parity(shape):
     w = torch.randn(Cout, 3, 3, Cin, device="cuda", generator=g,
                     dtype=torch.bfloat16) * (1.0 / (3 * Cin ** 0.5))
     b = torch.randn(Cout, device="cuda", generator=g) * 0.1
-    got = torch.ops.nvs3d.conv3x3_fwd(x, w, b)
+    got = torch.ops.nvs3d.conv3x3_fwd(x, w, b, None, 1.0)
     want = ref.frame_conv3x3(x.float(), w.float(), b.float())
     err = (got.float() - want).abs().max().item()
     scale = want.abs().max().item()
