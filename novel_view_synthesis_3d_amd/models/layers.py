@@ -138,10 +138,10 @@ class JointGroupNorm(nn.Module):
         self.scale = nn.Parameter(torch.ones(channels))
         self.bias = nn.Parameter(torch.zeros(channels))
 
-    def forward(self, x: torch.Tensor, film=None,
-                silu: bool = False) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, film=None, silu: bool = False,
+                p_drop: float = 0.0) -> torch.Tensor:
         return ops.joint_groupnorm(x, self.scale, self.bias, self.groups,
-                                   self.EPS, film, silu)
+                                   self.EPS, film, silu, p_drop)
 
     def flax_leaves(self):
         return [
@@ -218,9 +218,10 @@ class ResnetBlock(nn.Module):
             h_in = ops.avgpool_downsample2x(h_in)
         h = self.Conv_0(h)
         film = self.FiLM_0.packed(emb_silu)
-        h = self.GroupNorm_1(h, film=film, silu=True)
-        if self.dropout_rate > 0.0:
-            h = F.dropout(h, self.dropout_rate, training=self.training)
+        # the whole inter-conv segment is ONE kernel: GN+FiLM+SiLU+dropout
+        h = self.GroupNorm_1(h, film=film, silu=True,
+                             p_drop=(self.dropout_rate if self.training
+                                     else 0.0))
         if self.Dense_0 is not None:
             h_in = self.Dense_0(h_in)
         # residual + 1/sqrt(2) fused into the conv epilogue (xunet.py:92)

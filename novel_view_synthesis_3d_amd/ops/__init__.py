@@ -81,11 +81,12 @@ def frame_conv3x3(x, weight, bias, stride: int = 1):
 
 
 def joint_groupnorm(x, gamma, beta, groups: int, eps: float = 1e-6,
-                    film=None, silu: bool = False):
+                    film=None, silu: bool = False, p_drop: float = 0.0):
     if _use_hip(x, "joint_groupnorm"):
         return _HIP_MOD.joint_groupnorm(x, gamma, beta, groups, eps,
-                                        film, silu)
-    return ref.joint_groupnorm(x, gamma, beta, groups, eps, film, silu)
+                                        film, silu, p_drop)
+    return ref.joint_groupnorm(x, gamma, beta, groups, eps, film, silu,
+                               p_drop)
 
 
 def attention(q, k, v):

@@ -32,11 +32,24 @@ namespace {
 
 constexpr int MAX_GROUPS = 32;
 
+// Counter-based dropout mask: PCG hash of (element index, seed). The
+// backward REGENERATES the mask from the same seed — no mask tensor.
+__device__ __forceinline__ bool keep_elem(unsigned idx, unsigned seed,
+                                          unsigned thresh) {
+  unsigned v = idx * 0x9E3779B9u + seed;
+  v = v * 747796405u + 2891336453u;
+  unsigned w = ((v >> ((v >> 28) + 4u)) ^ v) * 277803737u;
+  return ((w >> 22) ^ w) >= thresh;
+}
+
 struct GnShape {
   int B, R, C, G, Cg;   // rows R = F*H*W; Cg = C/G
   int P;                // row chunks per batch
   int rowThreads;       // C / V
   int T;                // block threads = rowThreads * rowsPerIter
+  float drop_scale;     // 1/(1-p); 0 = no dropout
+  unsigned drop_seed;
+  unsigned drop_thresh; // p * 2^32
 };
 
 // ---------------------------------------------------------------------------
@@ -151,6 +164,10 @@ __global__ void gn_fwd_apply(const T* __restrict__ x,
       float u = (to_f32(px.v[j]) - mu) * r * gm[j] + bt[j];
       if (FILM) u = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       if (SILU) u = u * sigmoidf_fast(u);
+      if (s.drop_scale != 0.f) {
+        u = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+                ? u * s.drop_scale : 0.f;
+      }
       from_f32(u, po.v[j]);
     }
     pstore<T, V>(y + off, po);
@@ -223,6 +240,10 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
       float v = u;
       if (FILM) v = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       float dv = to_f32(pdy.v[j]);
+      if (s.drop_scale != 0.f) {
+        dv = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+                 ? dv * s.drop_scale : 0.f;
+      }
       if (SILU) {
         const float sg = sigmoidf_fast(v);
         dv *= sg * (1.f + v * (1.f - sg));
@@ -326,6 +347,10 @@ __global__ void gn_bwd_apply(const T* __restrict__ dy,
       float v = u;
       if (FILM) v = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       float dv = to_f32(pdy.v[j]);
+      if (s.drop_scale != 0.f) {
+        dv = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+                 ? dv * s.drop_scale : 0.f;
+      }
       if (SILU) {
         const float sg = sigmoidf_fast(v);
         dv *= sg * (1.f + v * (1.f - sg));
@@ -395,7 +420,8 @@ bool pick_block(GnShape& s, int V) {
 std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> film_in,
-                                  int64_t groups, double eps, bool silu) {
+                                  int64_t groups, double eps, bool silu,
+                                  double p_drop, int64_t drop_seed) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous CUDA");
   TORCH_CHECK(x.dim() == 5, "x must be (B,F,H,W,C)");
   const long B = x.size(0);
@@ -410,6 +436,9 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
   }
 
   GnShape s = make_shape(B, R, C, groups);
+  s.drop_scale = p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 0.f;
+  s.drop_seed = (unsigned)drop_seed;
+  s.drop_thresh = (unsigned)(p_drop * 4294967296.0);
   const int elem = x.scalar_type() == torch::kFloat ? 4 : 2;
   int V = pick_vec(s.Cg, elem);
   TORCH_CHECK(pick_block(s, V), "unsupported GN shape C=", C);
@@ -455,7 +484,8 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor beta,
                                   c10::optional<torch::Tensor> film_in,
                                   torch::Tensor mean, torch::Tensor rstd,
-                                  int64_t groups, bool silu) {
+                                  int64_t groups, bool silu,
+                                  double p_drop, int64_t drop_seed) {
   TORCH_CHECK(dy.is_cuda() && x.is_contiguous());
   auto dyc = dy.contiguous();
   const long B = x.size(0);
@@ -464,6 +494,9 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
   const bool film = film_in.has_value();
 
   GnShape s = make_shape(B, R, C, groups);
+  s.drop_scale = p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 0.f;
+  s.drop_seed = (unsigned)drop_seed;
+  s.drop_thresh = (unsigned)(p_drop * 4294967296.0);
   const int elem = x.scalar_type() == torch::kFloat ? 4 : 2;
   int V = pick_vec(s.Cg, elem);
   TORCH_CHECK(pick_block(s, V), "unsupported GN shape C=", C);

@@ -39,15 +39,17 @@ def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
 
 class _GnFused(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, film, groups, eps, silu):
+    def forward(ctx, x, gamma, beta, film, groups, eps, silu, p_drop, seed):
         x = x.contiguous()
         has_film = film is not None
         if has_film:
             film = film.to(x.dtype).contiguous()
-        y, mean, rstd = _OPS.gn_fwd(x, gamma, beta, film, groups, eps, silu)
+        y, mean, rstd = _OPS.gn_fwd(x, gamma, beta, film, groups, eps, silu,
+                                    p_drop, seed)
         ctx.save_for_backward(x, gamma, beta, mean, rstd,
                               *((film,) if has_film else ()))
         ctx.groups, ctx.silu, ctx.film = groups, silu, has_film
+        ctx.p_drop, ctx.seed = p_drop, seed
         return y
 
     @staticmethod
@@ -58,18 +60,23 @@ class _GnFused(torch.autograd.Function):
             x, gamma, beta, mean, rstd = ctx.saved_tensors
             film = None
         outs = _OPS.gn_bwd(dy, x, gamma, beta, film, mean, rstd,
-                           ctx.groups, ctx.silu)
+                           ctx.groups, ctx.silu, ctx.p_drop, ctx.seed)
         if ctx.film:
             dx, dgamma, dbeta, dfilm = outs
         else:
             dx, dgamma, dbeta = outs
             dfilm = None
         return (dx, dgamma.to(gamma.dtype), dbeta.to(beta.dtype),
-                dfilm, None, None, None)
+                dfilm, None, None, None, None, None)
 
 
-def joint_groupnorm(x, gamma, beta, groups, eps=1e-6, film=None, silu=False):
-    return _GnFused.apply(x, gamma, beta, film, groups, eps, silu)
+def joint_groupnorm(x, gamma, beta, groups, eps=1e-6, film=None, silu=False,
+                    p_drop=0.0):
+    """GroupNorm(+FiLM)(+SiLU)(+dropout) fully fused; the dropout mask is a
+    counter-based hash regenerated in backward (no mask tensor)."""
+    seed = int(torch.randint(0, 2 ** 31 - 1, (1,)).item()) if p_drop > 0 else 0
+    return _GnFused.apply(x, gamma, beta, film, groups, eps, silu,
+                          p_drop, seed)
 
 
 # ---------------------------------------------------------------------------

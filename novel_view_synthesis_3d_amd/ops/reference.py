@@ -55,7 +55,7 @@ def frame_conv3x3(x: torch.Tensor, weight: torch.Tensor,
 def joint_groupnorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                     groups: int, eps: float = 1e-6,
                     film: Optional[torch.Tensor] = None,
-                    silu: bool = False) -> torch.Tensor:
+                    silu: bool = False, p_drop: float = 0.0) -> torch.Tensor:
     """K3(+K5+K4): GroupNorm with statistics jointly over BOTH frames and all
     spatial positions per (batch, group) — the reference's frame-axis GroupNorm
     (/root/reference/model/xunet.py:46-52; flax GroupNorm reduces over all
@@ -79,6 +79,8 @@ def joint_groupnorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
         h = h * (1.0 + scale) + shift
     if silu:
         h = F.silu(h)
+    if p_drop > 0.0:
+        h = F.dropout(h, p_drop, training=True)
     return h.to(x.dtype)
 
 

