@@ -30,6 +30,11 @@ torch::Tensor attn_p_from_lse(torch::Tensor s, torch::Tensor lse,
                               double scale);
 torch::Tensor attn_ds(torch::Tensor p, torch::Tensor dp, torch::Tensor delta,
                       double scale);
+torch::Tensor up2x_fwd(torch::Tensor x);
+torch::Tensor up2x_bwd(torch::Tensor dout);
+torch::Tensor pool2x_fwd(torch::Tensor x);
+torch::Tensor pool2x_bwd(torch::Tensor dout);
+torch::Tensor add_scale(torch::Tensor a, torch::Tensor b, double scale);
 
 torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
                              torch::Tensor Kinv,
@@ -55,6 +60,11 @@ TORCH_LIBRARY(nvs3d, m) {
   m.def("im2col3x3(Tensor x, int stride, int m0, int m1, Tensor? out_buf) -> Tensor");
   m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");
   m.def("attn_ds(Tensor p, Tensor dp, Tensor delta, float scale) -> Tensor");
+  m.def("up2x_fwd(Tensor x) -> Tensor");
+  m.def("up2x_bwd(Tensor dout) -> Tensor");
+  m.def("pool2x_fwd(Tensor x) -> Tensor");
+  m.def("pool2x_bwd(Tensor dout) -> Tensor");
+  m.def("add_scale(Tensor a, Tensor b, float scale) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
@@ -67,4 +77,9 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("im2col3x3", im2col3x3);
   m.impl("attn_p_from_lse", attn_p_from_lse);
   m.impl("attn_ds", attn_ds);
+  m.impl("up2x_fwd", up2x_fwd);
+  m.impl("up2x_bwd", up2x_bwd);
+  m.impl("pool2x_fwd", pool2x_fwd);
+  m.impl("pool2x_bwd", pool2x_bwd);
+  m.impl("add_scale", add_scale);
 }

@@ -23,7 +23,8 @@ except (AttributeError, RuntimeError):
 _OPS = torch.ops.nvs3d
 
 # ops with a HIP implementation (consulted by ops/__init__.py dispatch)
-HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3", "attention"}
+HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3", "attention",
+       "nearest_upsample2x", "avgpool_downsample2x", "residual_scale_add"}
 
 
 def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
@@ -382,3 +383,65 @@ def frame_conv3x3_residual(x, weight, bias, residual, res_scale):
         return _FrameConv3x3.apply(x, weight, bias, residual, res_scale)
     y = frame_conv3x3(x, weight, bias, 1)
     return (y + residual) * res_scale
+
+
+# ---------------------------------------------------------------------------
+# Resampling (K8/K9) + residual (K10) elementwise kernels
+# ---------------------------------------------------------------------------
+
+class _Up2x(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        return _OPS.up2x_fwd(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _OPS.up2x_bwd(dy.contiguous())
+
+
+class _Pool2x(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        return _OPS.pool2x_fwd(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _OPS.pool2x_bwd(dy.contiguous())
+
+
+class _AddScale(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, scale):
+        ctx.scale = scale
+        return _OPS.add_scale(a.contiguous(), b.contiguous(), scale)
+
+    @staticmethod
+    def backward(ctx, dy):
+        g = dy * ctx.scale
+        return g, g, None
+
+
+def _elts_ok(x):
+    return x.size(-1) % 8 == 0 and x.dtype in (torch.bfloat16, torch.float32)
+
+
+def nearest_upsample2x(x):
+    if _elts_ok(x):
+        return _Up2x.apply(x)
+    from novel_view_synthesis_3d_amd.ops import reference as _ref
+    return _ref.nearest_upsample2x(x)
+
+
+def avgpool_downsample2x(x):
+    if _elts_ok(x):
+        return _Pool2x.apply(x)
+    from novel_view_synthesis_3d_amd.ops import reference as _ref
+    return _ref.avgpool_downsample2x(x)
+
+
+def residual_scale_add(h, h_in):
+    import math
+    if _elts_ok(h) and h.dtype == h_in.dtype:
+        return _AddScale.apply(h, h_in, 1.0 / math.sqrt(2.0))
+    from novel_view_synthesis_3d_amd.ops import reference as _ref
+    return _ref.residual_scale_add(h, h_in)

@@ -428,7 +428,10 @@ def test_gn_fused_dropout():
                                       p, 12345)
     y2, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
                                       p, 12345)
-    assert torch.equal(y1, y2)  # same seed -> same mask
+    # same seed -> same MASK (values can differ in the last ulp: the GN
+    # stats reduction uses LDS float atomics, so summation order varies)
+    assert torch.equal(y1 == 0, y2 == 0)
+    assert torch.allclose(y1.float(), y2.float(), atol=1e-2)
     mask = y1 != 0
     frac = mask.float().mean().item()
     assert abs(frac - (1 - p)) < 0.02, frac
@@ -438,3 +441,45 @@ def test_gn_fused_dropout():
     xx = x.detach().clone().requires_grad_(True)
     out = hip_ops.joint_groupnorm(xx, gm, bt, 32, 1e-6, None, True, 0.0)
     assert out.shape == y0.shape
+
+
+def test_resample_kernels_parity():
+    g = torch.Generator(device="cuda").manual_seed(8)
+    for dtype in (torch.bfloat16, torch.float32):
+        x = torch.randn(2, 2, 8, 8, 64, device="cuda", generator=g,
+                        dtype=dtype, requires_grad=True)
+        up = hip_ops.nearest_upsample2x(x)
+        want = ref.nearest_upsample2x(x.detach().float())
+        assert (up.float() - want).abs().max().item() < 1e-5
+        dy = torch.randn_like(up, dtype=torch.float32).to(dtype)
+        up.backward(dy)
+        # upsample bwd = 2x2 sum of dy
+        want_dx = dy.float().reshape(2, 2, 8, 2, 8, 2, 64).sum(dim=(3, 5))
+        assert (x.grad.float() - want_dx).abs().max().item() < 1e-2
+
+        x2 = torch.randn(2, 2, 8, 8, 64, device="cuda", generator=g,
+                         dtype=dtype, requires_grad=True)
+        dn = hip_ops.avgpool_downsample2x(x2)
+        want = ref.avgpool_downsample2x(x2.detach().float())
+        assert (dn.float() - want).abs().max().item() < 1e-2
+        dy2 = torch.randn_like(dn)
+        dn.backward(dy2)
+        want_dx2 = (dy2.float() / 4).reshape(2, 2, 4, 1, 4, 1, 64) \
+            .expand(2, 2, 4, 2, 4, 2, 64).reshape(2, 2, 8, 8, 64)
+        assert (x2.grad.float() - want_dx2).abs().max().item() < 1e-2
+
+
+def test_add_scale_kernel():
+    import math
+    g = torch.Generator(device="cuda").manual_seed(9)
+    a = torch.randn(2, 2, 8, 8, 64, device="cuda", generator=g,
+                    dtype=torch.bfloat16, requires_grad=True)
+    b = torch.randn_like(a, dtype=torch.float32).to(torch.bfloat16) \
+        .requires_grad_(True)
+    y = hip_ops.residual_scale_add(a, b)
+    want = (a.detach().float() + b.detach().float()) / math.sqrt(2)
+    assert (y.float() - want).abs().max().item() < 1e-2
+    y.sum().backward()
+    assert torch.allclose(a.grad.float(),
+                          torch.full_like(a.grad.float(),
+                                          1 / math.sqrt(2)), atol=1e-2)
