@@ -435,8 +435,9 @@ def test_gn_fused_dropout():
     mask = y1 != 0
     frac = mask.float().mean().item()
     assert abs(frac - (1 - p)) < 0.02, frac
-    kept = (y1[mask].float() - y0[mask].float() / (1 - p)).abs().max().item()
-    assert kept < 2e-2, kept
+    kept = (y1[mask].float() - y0[mask].float() / (1 - p)).abs()
+    rel = (kept / (y0[mask].float().abs() / (1 - p) + 1e-3)).max().item()
+    assert rel < 2e-2, rel  # bf16 ulp on both sides
     # backward: dx is zero-consistent with the same mask
     xx = x.detach().clone().requires_grad_(True)
     out = hip_ops.joint_groupnorm(xx, gm, bt, 32, 1e-6, None, True, 0.0)
@@ -466,7 +467,8 @@ def test_resample_kernels_parity():
         dn.backward(dy2)
         want_dx2 = (dy2.float() / 4).reshape(2, 2, 4, 1, 4, 1, 64) \
             .expand(2, 2, 4, 2, 4, 2, 64).reshape(2, 2, 8, 8, 64)
-        assert (x2.grad.float() - want_dx2).abs().max().item() < 1e-2
+        tol = 5e-2 if dtype == torch.bfloat16 else 1e-5
+        assert (x2.grad.float() - want_dx2).abs().max().item() < tol
 
 
 def test_add_scale_kernel():
