@@ -462,7 +462,8 @@ def test_resample_kernels_parity():
                          dtype=dtype, requires_grad=True)
         dn = hip_ops.avgpool_downsample2x(x2)
         want = ref.avgpool_downsample2x(x2.detach().float())
-        assert (dn.float() - want).abs().max().item() < 1e-2
+        tol = 3e-2 if dtype == torch.bfloat16 else 1e-5  # bf16 ulp at ~2
+        assert (dn.float() - want).abs().max().item() < tol
         dy2 = torch.randn_like(dn)
         dn.backward(dy2)
         want_dx2 = (dy2.float() / 4).reshape(2, 2, 4, 1, 4, 1, 64) \
