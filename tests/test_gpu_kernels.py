@@ -456,7 +456,8 @@ def test_resample_kernels_parity():
         up.backward(dy)
         # upsample bwd = 2x2 sum of dy
         want_dx = dy.float().reshape(2, 2, 8, 2, 8, 2, 64).sum(dim=(3, 5))
-        assert (x.grad.float() - want_dx).abs().max().item() < 1e-2
+        tol = 5e-2 if dtype == torch.bfloat16 else 1e-5  # bf16 ulp at ~4
+        assert (x.grad.float() - want_dx).abs().max().item() < tol
 
         x2 = torch.randn(2, 2, 8, 8, 64, device="cuda", generator=g,
                          dtype=dtype, requires_grad=True)
