@@ -244,6 +244,215 @@ void conv3x3_igemm(const bf16* __restrict__ x,   // (IMG,H,W,Cin)
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// 256x256-tile variant (8 waves as 2Mx4N, BK=64) — the higher-throughput
+// structure of cdna_hip_programming.md §5: halves the staging traffic per
+// FLOP and runs 64 MFMA per wave between barriers (vs 32 in the 128² kernel).
+// Per K-tile: 4 quadrant phases with no internal barriers (each wave reads
+// its own A/B halves); ONE vmcnt(0)+barrier per K-tile. A/B staged by glds
+// in 4 half-tiles (128 rows x 64 k each), double-buffered (4 x 16KB x 2 =
+// 128 KiB LDS), same source-side XOR swizzle and zero-buffer padding.
+// Selected when M % 256 == 0 and Cout % 256 == 0 (L0-L2 of the full model).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int BM2 = 256;
+constexpr int BN2 = 256;
+constexpr int T2 = 512;
+
+__global__ __launch_bounds__(T2)
+void conv3x3_igemm_256(const bf16* __restrict__ x,
+                       const bf16* __restrict__ w,   // (Cout, 9*Cin)
+                       const float* __restrict__ bias,
+                       const bf16* __restrict__ zbuf,
+                       const bf16* __restrict__ residual,
+                       float out_scale,
+                       bf16* __restrict__ out,
+                       ConvShape s, int nblocks_m) {
+  int bid = blockIdx.x;
+  const int nwg = nblocks_m * (s.Cout / BN2);
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int bm = bid % nblocks_m;
+  const int bn = bid / nblocks_m;
+  const int m0 = bm * BM2;
+  const int n0 = bn * BN2;
+
+  // LDS: [buf(2)][operand A=0/B=1][half(2)][16KB]
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto half_base = [&](int buf, int op, int half) -> char* {
+    return smem + (((buf * 2 + op) * 2 + half) << 14);
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;   // 0..7
+  const int wm = wave >> 2;    // A half this wave consumes
+  const int wn = wave & 3;     // B cols: half wn>>1, sub (wn&1)*64
+
+  // glds slot decode: wave issues 2 glds per half-tile; slot s covers LDS
+  // bytes [(wave*2+s)*1024, +1024) of that half (lane-linear). Element under
+  // the ((row&7)<<4) XOR swizzle: row = o>>7, kp = ((o ^ ((row&7)<<4)) & 127) >> 4.
+  int arow_h[2][2], arow_w[2][2], arow_ok[2][2], kp_[2];  // [half][slot]
+  long arow_base[2][2];
+  int bco[2][2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+#pragma unroll
+    for (int slot = 0; slot < 2; ++slot) {
+      const int o = (wave * 2 + slot) * 1024 + lane * 16;
+      const int row = o >> 7;             // 0..127 within the half
+      kp_[slot] = ((o ^ ((row & 7) << 4)) & 127) >> 4;
+      const int m = m0 + half * 128 + row;
+      arow_w[half][slot] = m % s.W;
+      arow_h[half][slot] = (m / s.W) % s.H;
+      arow_ok[half][slot] = m < s.M;
+      arow_base[half][slot] = ((long)(m / (s.W * s.H)) * s.H) * s.W * s.Cin;
+      bco[half][slot] = n0 + half * 128 + row;
+    }
+  }
+
+  auto stage_tile = [&](int kstep, int buf) {
+    const int plane = kstep / s.steps_per_plane;
+    const int ci0 = (kstep % s.steps_per_plane) * BK;
+    const int dy = plane / 3 - 1;
+    const int dx = plane % 3 - 1;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+#pragma unroll
+      for (int slot = 0; slot < 2; ++slot) {
+        const int hh = arow_h[half][slot] + dy;
+        const int ww = arow_w[half][slot] + dx;
+        const bool valid = arow_ok[half][slot] & (hh >= 0) & (hh < s.H)
+                           & (ww >= 0) & (ww < s.W);
+        const bf16* srcA = valid
+            ? x + arow_base[half][slot] + ((long)hh * s.W + ww) * s.Cin
+                + ci0 + kp_[slot] * 8
+            : zbuf;
+        __builtin_amdgcn_global_load_lds(as_global(srcA),
+            as_shared(half_base(buf, 0, half) + (wave * 2 + slot) * 1024),
+            16, 0, 0);
+        const bf16* srcB = w + (long)bco[half][slot] * (9 * s.Cin)
+                           + kstep * BK + kp_[slot] * 8;
+        __builtin_amdgcn_global_load_lds(as_global(srcB),
+            as_shared(half_base(buf, 1, half) + (wave * 2 + slot) * 1024),
+            16, 0, 0);
+      }
+    }
+  };
+
+  float acc[8][4][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  stage_tile(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < s.ksteps; ++t) {
+    const int cur = t & 1;
+    const char* baseA = half_base(cur, 0, wm);
+    const char* baseB = half_base(cur, 1, wn >> 1);
+    // B fragments for the whole K-tile (shared by all 4 quadrants)
+    bf16x8 bfr[4][2];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int row = (wn & 1) * 64 + j * 16 + (lane & 15);
+        bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
+            baseB + swz(row, kk * 32 + (lane >> 4) * 8));
+      }
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      if (q == 0 && t + 1 < s.ksteps) {
+        stage_tile(t + 1, cur ^ 1);  // overlaps the whole tile's compute
+      }
+      bf16x8 afr[2][2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int row = q * 32 + i * 16 + (lane & 15);
+          afr[i][kk] = *reinterpret_cast<const bf16x8*>(
+              baseA + swz(row, kk * 32 + (lane >> 4) * 8));
+        }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]) =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afr[i][kk], bfr[j][kk],
+                    *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]), 0, 0, 0);
+          }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  // epilogue: stage the 256x256 bf16 tile through LDS (128 KiB) and store
+  // coalesced, with fused bias (+ residual)*scale
+  bf16* ldsC = reinterpret_cast<bf16*>(smem);
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int col = wn * 64 + j * 16 + (lane & 15);
+    const float bj = bias != nullptr ? bias[n0 + col] : 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int rbase = wm * 128 + i * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        ldsC[(rbase + r) * BN2 + col] = __float2bfloat16(acc[i][j][r] + bj);
+      }
+    }
+  }
+  __syncthreads();
+  {
+    // 256 rows x 32 packs = 8192 packs / 512 threads = 16 each
+#pragma unroll
+    for (int it = 0; it < 16; ++it) {
+      const int p = tid + it * T2;
+      const int row = p >> 5;
+      const int cp = p & 31;
+      const int m = m0 + row;
+      if (m < s.M) {
+        Pack<bf16, 8> v = *reinterpret_cast<Pack<bf16, 8>*>(
+            ldsC + row * BN2 + cp * 8);
+        const long goff = (long)m * s.Cout + n0 + cp * 8;
+        if (residual != nullptr) {
+          Pack<bf16, 8> rv = pload<bf16, 8>(residual + goff);
+#pragma unroll
+          for (int jj = 0; jj < 8; ++jj) {
+            from_f32((to_f32(v.v[jj]) + to_f32(rv.v[jj])) * out_scale,
+                     v.v[jj]);
+          }
+        } else if (out_scale != 1.0f) {
+#pragma unroll
+          for (int jj = 0; jj < 8; ++jj) {
+            from_f32(to_f32(v.v[jj]) * out_scale, v.v[jj]);
+          }
+        }
+        pstore<bf16, 8>(out + goff, v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias,
                           c10::optional<torch::Tensor> residual,
@@ -277,15 +486,32 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
     zbuf = torch::zeros({64}, x.options());
   }
 
-  const int nblocks_m = (s.M + BM - 1) / BM;
-  const int grid = nblocks_m * (s.Cout / BN);
-  const size_t lds = 2 * (BM + BN) * BK * 2;  // 64 KB
   auto stream = at::hip::getCurrentHIPStream();
   if (residual.has_value()) {
     TORCH_CHECK(residual->is_contiguous()
                 && residual->sizes() == out.sizes()
                 && residual->scalar_type() == torch::kBFloat16);
   }
+  // 256x256-tile variant where it fills the chip (L0-L2 full-config convs)
+  const bool use256 = (s.M % 256 == 0) && (s.Cout % 256 == 0)
+      && ((long)(s.M / 256) * (s.Cout / 256) >= 128);
+  if (use256) {
+    const int nb_m = s.M / BM2;
+    hipLaunchKernelGGL(conv3x3_igemm_256, dim3(nb_m * (s.Cout / BN2)),
+        dim3(T2), 128 * 1024, stream,
+        reinterpret_cast<const bf16*>(x.data_ptr()),
+        reinterpret_cast<const bf16*>(w.data_ptr()),
+        bias.has_value() ? biasf.data_ptr<float>() : nullptr,
+        reinterpret_cast<const bf16*>(zbuf.data_ptr()),
+        residual.has_value()
+            ? reinterpret_cast<const bf16*>(residual->data_ptr()) : nullptr,
+        (float)out_scale,
+        reinterpret_cast<bf16*>(out.data_ptr()), s, nb_m);
+    return out;
+  }
+  const int nblocks_m = (s.M + BM - 1) / BM;
+  const int grid = nblocks_m * (s.Cout / BN);
+  const size_t lds = 2 * (BM + BN) * BK * 2;  // 64 KB
   hipLaunchKernelGGL(conv3x3_igemm, dim3(grid), dim3(THREADS), lds, stream,
       reinterpret_cast<const bf16*>(x.data_ptr()),
       reinterpret_cast<const bf16*>(w.data_ptr()),
