@@ -42,7 +42,8 @@ def bench_conv():
                         dtype=torch.bfloat16) * 0.02
         b = torch.randn(Co, device="cuda")
         flops = 2.0 * B * F * H * W * 9 * Ci * Co
-        t_hip = timeit(lambda: torch.ops.nvs3d.conv3x3_fwd(x, w, b))
+        t_hip = timeit(
+            lambda: torch.ops.nvs3d.conv3x3_fwd(x, w, b, None, 1.0))
         t_mio = timeit(lambda: ref.frame_conv3x3(x, w, b.to(torch.bfloat16)))
         rec = {"op": "conv3x3_fwd", "shape": [B, F, H, W, Ci, Co],
                "hip_ms": round(t_hip, 3), "miopen_ms": round(t_mio, 3),
