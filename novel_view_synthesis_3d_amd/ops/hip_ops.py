@@ -223,18 +223,6 @@ def _wgrad_im2col_gemm(x, w, dy, has_bias):
     return dw, db
 
 
-def _miopen_wgrad(x, w, dy, has_bias):
-    xg = _nchw_view(x)
-    dyg = _nchw_view(dy)
-    wg = w.permute(0, 3, 1, 2)  # (Cout,Cin,3,3) channels_last view
-    _, dw, db = torch.ops.aten.convolution_backward(
-        dyg, xg, wg, [w.shape[0]] if has_bias else None,
-        [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-        [False, True, has_bias])
-    dw_ohwi = dw.permute(0, 2, 3, 1).contiguous()
-    return dw_ohwi, db
-
-
 def _pad_c8(t):
     """Pad the channel (last) dim to a multiple of 8 (im2col needs 16B packs)."""
     c = t.shape[-1]
