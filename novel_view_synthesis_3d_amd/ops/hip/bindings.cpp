@@ -24,8 +24,9 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           double out_scale);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
-torch::Tensor im2col3x3(torch::Tensor x, int64_t stride, int64_t m0,
-                        int64_t m1, c10::optional<torch::Tensor> out_buf);
+torch::Tensor im2col3x3(torch::Tensor x, int64_t stride, int64_t nplanes,
+                        int64_t m0, int64_t m1,
+                        c10::optional<torch::Tensor> out_buf);
 torch::Tensor attn_p_from_lse(torch::Tensor s, torch::Tensor lse,
                               double scale);
 torch::Tensor attn_ds(torch::Tensor p, torch::Tensor dp, torch::Tensor delta,
@@ -57,7 +58,7 @@ TORCH_LIBRARY(nvs3d, m) {
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias, Tensor? residual, float out_scale) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
-  m.def("im2col3x3(Tensor x, int stride, int m0, int m1, Tensor? out_buf) -> Tensor");
+  m.def("im2col3x3(Tensor x, int stride, int nplanes, int m0, int m1, Tensor? out_buf) -> Tensor");
   m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");
   m.def("attn_ds(Tensor p, Tensor dp, Tensor delta, float scale) -> Tensor");
   m.def("up2x_fwd(Tensor x) -> Tensor");

@@ -198,27 +198,44 @@ def _miopen_conv(x, w, bias):
     return out.reshape(*shape[:-1], w.shape[0]).contiguous()
 
 
-_WGRAD_CHUNK_BYTES = 96 * 1024 * 1024  # keep the im2col slab L3-resident
-
-
 def _wgrad_im2col_gemm(x, w, dy, has_bias):
-    """Conv wgrad as CHUNKED im2col + hipBLASLt GEMMs: dW = sum_c A_c^T dy_c.
-    Chunks are sized so the im2col slab stays inside the 256 MiB Infinity
-    Cache — the GEMM re-reads it from L3, not HBM. Beats MIOpen's wrw
-    kernels on gfx950 (profiles/)."""
+    """Conv wgrad via the 3-row-shift decomposition: materialize only the
+    ROW-shifted im2col A3[m][(dy,ci)] (3x the input instead of 9x); each
+    kernel COLUMN dx is then one hipBLASLt GEMM between flat-shifted views:
+
+        dW[:, dx] = A3[m-(dx-1)]^T @ dyf[m]  -  wrap corrections
+
+    The flat +-1 shift wraps across image-row boundaries exactly at the
+    rows with w==0 (dx=0) / w==W-1 (dx=2); the true contribution there is
+    zero (SAME padding), so those rows' outer products (M/W rows) are
+    gathered and subtracted — exact, tested against the fp32 oracle."""
     cout, _, _, cin = w.shape
     M = dy.numel() // cout
+    W = x.shape[-2]
     dyf = dy.reshape(M, cout)
-    rows = max(4096, _WGRAD_CHUNK_BYTES // (18 * cin))  # bf16, 9*Cin cols
-    rows = min(rows, M)
-    buf = torch.empty(rows, 9 * cin, device=x.device, dtype=x.dtype)
-    dw32 = None
-    for m0 in range(0, M, rows):
-        m1 = min(M, m0 + rows)
-        A = _OPS.im2col3x3(x, 1, m0, m1, buf)
-        part = torch.matmul(A.transpose(0, 1), dyf[m0:m1])
-        dw32 = part.float() if dw32 is None else dw32.add_(part.float())
-    dw = dw32.to(x.dtype).reshape(3, 3, cin, cout).permute(3, 0, 1, 2)         .contiguous()
+    A3 = _OPS.im2col3x3(x, 1, 3, 0, -1, None)        # (M, 3*Cin)
+    dev = x.device
+
+    blocks = []
+    # dx = 0: pair A3[m-1] with dyf[m]
+    g0 = torch.matmul(A3[:M - 1].transpose(0, 1), dyf[1:]).float()
+    idx0 = torch.arange(W, M, W, device=dev)          # w==0, m>=1
+    g0 -= torch.matmul(A3.index_select(0, idx0 - 1).transpose(0, 1),
+                       dyf.index_select(0, idx0)).float()
+    blocks.append(g0)
+    # dx = 1: aligned
+    blocks.append(torch.matmul(A3.transpose(0, 1), dyf).float())
+    # dx = 2: pair A3[m+1] with dyf[m]
+    g2 = torch.matmul(A3[1:].transpose(0, 1), dyf[:M - 1]).float()
+    idx2 = torch.arange(W - 1, M - 1, W, device=dev)  # w==W-1, m<M-1
+    g2 -= torch.matmul(A3.index_select(0, idx2 + 1).transpose(0, 1),
+                       dyf.index_select(0, idx2)).float()
+    blocks.append(g2)
+
+    # blocks[dx]: (3*Cin, Cout), rows = (dy, ci) -> dW OHWI (Cout,3,3,Cin)
+    dwf = torch.stack(blocks, dim=1)                  # (3*Cin, 3dx, Cout)
+    dw = (dwf.reshape(3, cin, 3, cout).permute(3, 0, 2, 1)
+          .contiguous().to(x.dtype))
     db = dyf.sum(dim=0, dtype=torch.float32).to(dy.dtype) if has_bias else None
     return dw, db
 
@@ -239,7 +256,7 @@ def _gemm_conv_fwd(x, w, bias, stride):
     im2col kernel + ONE hipBLASLt GEMM. Replaces MIOpen entirely."""
     cout, _, _, cin = w.shape
     xp, _ = _pad_c8(x.contiguous())
-    A = _OPS.im2col3x3(xp, stride, 0, -1, None)          # (Mo, 9*Cpad)
+    A = _OPS.im2col3x3(xp, stride, 9, 0, -1, None)          # (Mo, 9*Cpad)
     cpad = xp.shape[-1]
     w2 = w.permute(1, 2, 3, 0)                            # (3,3,Cin,Cout)
     if cpad != cin:
@@ -287,7 +304,7 @@ class _FrameConvGeneric(torch.autograd.Function):
             dyf = dy.reshape(-1, cout)
             xp, _ = _pad_c8(x)
             cpad = xp.shape[-1]
-            A = _OPS.im2col3x3(xp, ctx.stride, 0, -1, None)
+            A = _OPS.im2col3x3(xp, ctx.stride, 9, 0, -1, None)
             dwf = torch.matmul(A.transpose(0, 1), dyf)    # (9*Cpad, Cout)
             dw = dwf.reshape(3, 3, cpad, cout)[:, :, :cin, :]                 .permute(3, 0, 1, 2).contiguous()
             if not ctx.needs_input_grad[1]:
