@@ -27,8 +27,10 @@ def variance_scaling_(tensor: torch.Tensor, fan_in: int, scale: float = 1.0,
             return tensor.zero_()
     std = math.sqrt(scale / fan_in)
     with torch.no_grad():
-        tensor.normal_(0.0, 1.0, generator=generator)
-        tensor.clamp_(-2.0, 2.0)  # cheap truncation; matches jax to within tail mass
+        # true truncated normal (inverse-CDF, like jax truncated_normal) —
+        # clamping instead would pile mass at +-2 and overshoot the std ~9%
+        torch.nn.init.trunc_normal_(tensor, mean=0.0, std=1.0,
+                                    a=-2.0, b=2.0, generator=generator)
         tensor.mul_(std / _TRUNC_STD)
     return tensor
 
