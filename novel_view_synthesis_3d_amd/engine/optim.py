@@ -23,8 +23,15 @@ class FusedAdam(torch.optim.Optimizer):
         try:
             from novel_view_synthesis_3d_amd.ops import hip_ops
             return hip_ops
-        except Exception:
-            return None
+        except Exception as e:
+            import os
+            if os.environ.get("NVS3D_ALLOW_EAGER_GPU", "0") == "1":
+                return None
+            # same fail-loudly policy as ops dispatch: a GPU box must not
+            # silently fall back to the slow eager Adam
+            raise RuntimeError(
+                f"FusedAdam on GPU but nvs3d_hip extension unavailable: {e}"
+            ) from e
 
     @torch.no_grad()
     def step(self, closure=None):
