@@ -200,6 +200,11 @@ class Trainer:
     # gaps). Gradients land in stable p.grad storage; the bucketed RCCL
     # all-reduce and the fused Adam update run after each replay (RCCL
     # collectives are kept outside the capture).
+    # CAVEAT (why this is off by default besides being measured ~4% slower:
+    # the step is GPU-bound): the fused-GN dropout seed is drawn on the host
+    # per call, so under replay the dropout mask is frozen across steps —
+    # the same trace-time-freezing the reference suffers from (D2). Philox
+    # RNG ops (noise, cond_mask) DO advance correctly under replay.
 
     def _graph_body(self):
         if self.data_mode == "synthetic":
