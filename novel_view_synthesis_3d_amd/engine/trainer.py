@@ -244,7 +244,11 @@ class Trainer:
                 self.opt.step()
         torch.cuda.current_stream().wait_stream(s)
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        # capture on the SAME stream the warmup ran on: the parameters' grad
+        # accumulators were created on `s` during warmup, and accumulation
+        # escapes the capture (-> NaN on replay) if the capture stream
+        # differs from theirs.
+        with torch.cuda.graph(self._graph, stream=s):
             self._static_loss = self._graph_body()
         self.graph_active = True
         if self.is_main:
