@@ -106,13 +106,6 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     torch.cuda.synchronize()
 
 
-@pytest.mark.xfail(
-    reason="experimental (off by default): warmup-created AccumulateGrad "
-           "nodes sit on a side stream, so the in-graph backward's grad "
-           "accumulation escapes the capture -> params NaN after the first "
-           "optimizer step. Diagnosis + fix plan in NOTES_ROUND2.md; the "
-           "sampler graph path (no autograd) is unaffected and tested above.",
-    strict=False)
 def test_trainer_graph_mode(tmp_path):
     """hipGraph-captured training step: captures, replays, finite losses."""
     from novel_view_synthesis_3d_amd.config import TrainConfig, XUNetConfig
