@@ -132,7 +132,9 @@ class SceneClassDataset(torch.utils.data.Dataset):
                  cpu_noising: bool = False):
         self.samples_per_instance = samples_per_instance
         self.instance_dirs = sorted(glob(os.path.join(root_dir, "*/")))
-        assert len(self.instance_dirs) != 0, "No objects in the data directory"
+        if not self.instance_dirs:
+            raise FileNotFoundError(
+                f"no instance directories found under {root_dir!r}")
         if max_num_instances != -1:
             self.instance_dirs = self.instance_dirs[:max_num_instances]
 
@@ -154,11 +156,12 @@ class SceneClassDataset(torch.utils.data.Dataset):
         return int(np.sum(self.num_per_instance_observations))
 
     def get_instance_idx(self, idx: int):
-        obj_idx = 0
-        while idx >= 0:
-            idx -= self.num_per_instance_observations[obj_idx]
-            obj_idx += 1
-        return obj_idx - 1, int(idx + self.num_per_instance_observations[obj_idx - 1])
+        """Map a flat sample index to (instance index, observation index)
+        via the cumulative observation counts."""
+        bounds = np.cumsum(self.num_per_instance_observations)
+        obj_idx = int(np.searchsorted(bounds, idx, side="right"))
+        start = 0 if obj_idx == 0 else int(bounds[obj_idx - 1])
+        return obj_idx, int(idx - start)
 
     def collate_fn(self, batch_list):
         """Stack list-of-(observations, ground_truth) into tensor dicts.

@@ -158,13 +158,19 @@ class DDPMSampler:
         """Capture one sampler step as a hipGraph and replay it num_steps
         times. Per-step state (z, idx) lives in the captured buffers; RNG is
         graph-safe (torch captures the philox offset)."""
+        # warm up at most num_steps steps: indexing past the step tables is a
+        # device-side assert, and with nothing left to replay the graph is
+        # pointless anyway
+        n_warm = min(2, self.num_steps)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):  # warmup, required before capture
-            for _ in range(2):
+            for _ in range(n_warm):
                 self._step(z, idx, cond2, mask, tab, pose_cache)
         torch.cuda.current_stream().wait_stream(s)
         done_warmup = int(idx.item())
+        if self.num_steps - done_warmup <= 0:
+            return
 
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):

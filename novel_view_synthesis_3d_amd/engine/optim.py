@@ -52,7 +52,14 @@ class FusedAdam(torch.optim.Optimizer):
                     st["exp_avg_sq"] = torch.zeros_like(p)
             for p in params:
                 self.state[p]["step"] += 1
-            step = int(self.state[params[0]]["step"].item())
+            steps = {int(self.state[p]["step"].item()) for p in params}
+            # the fused plan applies ONE bias correction to the whole group;
+            # a param that skipped steps (no grad some iterations) would need
+            # per-param correction — fail loudly instead of silently drifting
+            assert len(steps) == 1, (
+                f"FusedAdam group {gi} has divergent per-param step counts "
+                f"{sorted(steps)}; per-group fused bias correction is invalid")
+            step = steps.pop()
             b1, b2 = group["betas"]
 
             hip = self._hip() if params[0].is_cuda else None

@@ -5,13 +5,15 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> film,
                                   int64_t groups, double eps, bool silu,
-                                  double p_drop, int64_t drop_seed);
+                                  double p_drop,
+                                  c10::optional<torch::Tensor> drop_seed);
 std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor gamma, torch::Tensor beta,
                                   c10::optional<torch::Tensor> film,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   int64_t groups, bool silu,
-                                  double p_drop, int64_t drop_seed);
+                                  double p_drop,
+                                  c10::optional<torch::Tensor> drop_seed);
 torch::Tensor rays_posenc(torch::Tensor R, torch::Tensor t, torch::Tensor Kinv,
                           c10::optional<torch::Tensor> mask,
                           int64_t H, int64_t W, torch::ScalarType out_dtype);
@@ -47,10 +49,10 @@ torch::Tensor rays_posenc_py(torch::Tensor R, torch::Tensor t,
 TORCH_LIBRARY(nvs3d, m) {
   m.def("gn_fwd(Tensor x, Tensor gamma, Tensor beta, Tensor? film, "
         "int groups, float eps, bool silu, float p_drop, "
-        "int drop_seed) -> Tensor[]");
+        "Tensor? drop_seed) -> Tensor[]");
   m.def("gn_bwd(Tensor dy, Tensor x, Tensor gamma, Tensor beta, "
         "Tensor? film, Tensor mean, Tensor rstd, "
-        "int groups, bool silu, float p_drop, int drop_seed) -> Tensor[]");
+        "int groups, bool silu, float p_drop, Tensor? drop_seed) -> Tensor[]");
   m.def("rays_posenc(Tensor R, Tensor t, Tensor Kinv, Tensor? mask, "
         "int H, int W, Tensor dtype_like) -> Tensor");
   m.def("fused_adam(Tensor ptrs, Tensor chunk_tensor, Tensor chunk_off, "

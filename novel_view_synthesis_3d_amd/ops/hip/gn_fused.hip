@@ -34,9 +34,12 @@ constexpr int MAX_GROUPS = 32;
 
 // Counter-based dropout mask: PCG hash of (element index, seed). The
 // backward REGENERATES the mask from the same seed — no mask tensor.
-__device__ __forceinline__ bool keep_elem(unsigned idx, unsigned seed,
-                                          unsigned thresh) {
-  unsigned v = idx * 0x9E3779B9u + seed;
+// Both 32-bit halves of the element offset feed the hash, so tensors with
+// >= 2^32 elements don't silently reuse mask values across the wrap.
+__device__ __forceinline__ bool keep_elem(unsigned long long idx,
+                                          unsigned seed, unsigned thresh) {
+  unsigned v = (unsigned)idx * 0x9E3779B9u
+             + (unsigned)(idx >> 32) * 0x85EBCA6Bu + seed;
   v = v * 747796405u + 2891336453u;
   unsigned w = ((v >> ((v >> 28) + 4u)) ^ v) * 277803737u;
   return ((w >> 22) ^ w) >= thresh;
@@ -48,7 +51,10 @@ struct GnShape {
   int rowThreads;       // C / V
   int T;                // block threads = rowThreads * rowsPerIter
   float drop_scale;     // 1/(1-p); 0 = no dropout
-  unsigned drop_seed;
+  // device-side seed (1-elem int32 tensor): read at kernel execution time,
+  // not baked in at launch, so dropout masks keep advancing under hipGraph
+  // replay (the host-drawn-seed D2-style freeze is gone)
+  const unsigned* __restrict__ seed_ptr;
   unsigned drop_thresh; // p * 2^32
 };
 
@@ -140,6 +146,7 @@ __global__ void gn_fwd_apply(const T* __restrict__ x,
   const int g = c0 / s.Cg;
   const float mu = lmean[g];
   const float r = lrstd[g];
+  const unsigned seed = s.drop_scale != 0.f ? *s.seed_ptr : 0u;
 
   float gm[V], bt[V];
 #pragma unroll
@@ -165,7 +172,7 @@ __global__ void gn_fwd_apply(const T* __restrict__ x,
       if (FILM) u = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       if (SILU) u = u * sigmoidf_fast(u);
       if (s.drop_scale != 0.f) {
-        u = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+        u = keep_elem(off + j, seed, s.drop_thresh)
                 ? u * s.drop_scale : 0.f;
       }
       from_f32(u, po.v[j]);
@@ -209,6 +216,7 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
   const int g = c0 / s.Cg;
   const float mu = mean[(size_t)b * s.G + g];
   const float r = rstd[(size_t)b * s.G + g];
+  const unsigned seed = s.drop_scale != 0.f ? *s.seed_ptr : 0u;
 
   float gm[V], bt[V], dgm[V], dbt[V];
 #pragma unroll
@@ -241,7 +249,7 @@ __global__ void gn_bwd_partials(const T* __restrict__ dy,
       if (FILM) v = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       float dv = to_f32(pdy.v[j]);
       if (s.drop_scale != 0.f) {
-        dv = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+        dv = keep_elem(off + j, seed, s.drop_thresh)
                  ? dv * s.drop_scale : 0.f;
       }
       if (SILU) {
@@ -320,6 +328,7 @@ __global__ void gn_bwd_apply(const T* __restrict__ dy,
   const float r = rstd[(size_t)b * s.G + g];
   const float m1 = lS1[g];
   const float m2 = lS2[g];
+  const unsigned seed = s.drop_scale != 0.f ? *s.seed_ptr : 0u;
 
   float gm[V], bt[V];
 #pragma unroll
@@ -348,7 +357,7 @@ __global__ void gn_bwd_apply(const T* __restrict__ dy,
       if (FILM) v = u * (1.f + to_f32(ps.v[j])) + to_f32(pt.v[j]);
       float dv = to_f32(pdy.v[j]);
       if (s.drop_scale != 0.f) {
-        dv = keep_elem((unsigned)(off + j), s.drop_seed, s.drop_thresh)
+        dv = keep_elem(off + j, seed, s.drop_thresh)
                  ? dv * s.drop_scale : 0.f;
       }
       if (SILU) {
@@ -417,11 +426,21 @@ bool pick_block(GnShape& s, int V) {
 // ---------------------------------------------------------------------------
 // ATen entry points
 // ---------------------------------------------------------------------------
+static const unsigned* seed_ptr_of(const c10::optional<torch::Tensor>& seed,
+                                   double p_drop) {
+  if (!(p_drop > 0)) return nullptr;
+  TORCH_CHECK(seed.has_value(), "p_drop > 0 needs a device seed tensor");
+  TORCH_CHECK(seed->is_cuda() && seed->scalar_type() == torch::kInt &&
+              seed->numel() == 1, "seed must be a 1-elem int32 CUDA tensor");
+  return reinterpret_cast<const unsigned*>(seed->data_ptr<int32_t>());
+}
+
 std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta,
                                   c10::optional<torch::Tensor> film_in,
                                   int64_t groups, double eps, bool silu,
-                                  double p_drop, int64_t drop_seed) {
+                                  double p_drop,
+                                  c10::optional<torch::Tensor> drop_seed) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous CUDA");
   TORCH_CHECK(x.dim() == 5, "x must be (B,F,H,W,C)");
   const long B = x.size(0);
@@ -437,7 +456,7 @@ std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
 
   GnShape s = make_shape(B, R, C, groups);
   s.drop_scale = p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 0.f;
-  s.drop_seed = (unsigned)drop_seed;
+  s.seed_ptr = seed_ptr_of(drop_seed, p_drop);
   s.drop_thresh = (unsigned)(p_drop * 4294967296.0);
   const int elem = x.scalar_type() == torch::kFloat ? 4 : 2;
   int V = pick_vec(s.Cg, elem);
@@ -485,7 +504,8 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> film_in,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   int64_t groups, bool silu,
-                                  double p_drop, int64_t drop_seed) {
+                                  double p_drop,
+                                  c10::optional<torch::Tensor> drop_seed) {
   TORCH_CHECK(dy.is_cuda() && x.is_contiguous());
   auto dyc = dy.contiguous();
   const long B = x.size(0);
@@ -495,7 +515,7 @@ std::vector<torch::Tensor> gn_bwd(torch::Tensor dy, torch::Tensor x,
 
   GnShape s = make_shape(B, R, C, groups);
   s.drop_scale = p_drop > 0 ? (float)(1.0 / (1.0 - p_drop)) : 0.f;
-  s.drop_seed = (unsigned)drop_seed;
+  s.seed_ptr = seed_ptr_of(drop_seed, p_drop);
   s.drop_thresh = (unsigned)(p_drop * 4294967296.0);
   const int elem = x.scalar_type() == torch::kFloat ? 4 : 2;
   int V = pick_vec(s.Cg, elem);

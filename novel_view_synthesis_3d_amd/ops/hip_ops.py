@@ -47,21 +47,21 @@ class _GnFused(torch.autograd.Function):
             film = film.to(x.dtype).contiguous()
         y, mean, rstd = _OPS.gn_fwd(x, gamma, beta, film, groups, eps, silu,
                                     p_drop, seed)
-        ctx.save_for_backward(x, gamma, beta, mean, rstd,
-                              *((film,) if has_film else ()))
+        extras = ([film] if has_film else []) + ([seed] if seed is not None
+                                                 else [])
+        ctx.save_for_backward(x, gamma, beta, mean, rstd, *extras)
         ctx.groups, ctx.silu, ctx.film = groups, silu, has_film
-        ctx.p_drop, ctx.seed = p_drop, seed
+        ctx.p_drop, ctx.has_seed = p_drop, seed is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        if ctx.film:
-            x, gamma, beta, mean, rstd, film = ctx.saved_tensors
-        else:
-            x, gamma, beta, mean, rstd = ctx.saved_tensors
-            film = None
+        saved = list(ctx.saved_tensors)
+        seed = saved.pop() if ctx.has_seed else None
+        film = saved.pop() if ctx.film else None
+        x, gamma, beta, mean, rstd = saved
         outs = _OPS.gn_bwd(dy, x, gamma, beta, film, mean, rstd,
-                           ctx.groups, ctx.silu, ctx.p_drop, ctx.seed)
+                           ctx.groups, ctx.silu, ctx.p_drop, seed)
         if ctx.film:
             dx, dgamma, dbeta, dfilm = outs
         else:
@@ -71,11 +71,34 @@ class _GnFused(torch.autograd.Function):
                 dfilm, None, None, None, None, None)
 
 
+_GN_SEED_CTR: dict = {}
+
+
+def _next_gn_seed(device: torch.device) -> torch.Tensor:
+    """Per-call dropout seed SNAPSHOT from a device-resident counter.
+
+    The clone + increment are device ops, so under hipGraph capture they are
+    recorded and each replay advances the counter and re-snapshots — dropout
+    masks keep changing across graph replays (the host-drawn-seed freeze the
+    round-1 review flagged is gone). Counter init happens eagerly (warmup
+    always precedes capture)."""
+    idx = device.index if device.index is not None else 0
+    ctr = _GN_SEED_CTR.get(idx)
+    if ctr is None:
+        init = int(torch.randint(0, 2 ** 30, (1,)).item())
+        ctr = torch.full((1,), init, dtype=torch.int32, device=device)
+        _GN_SEED_CTR[idx] = ctr
+    seed = ctr.clone()
+    # large odd stride decorrelates consecutive calls' hash streams
+    ctr.add_(2654435761 & 0x7FFFFFFF)
+    return seed
+
+
 def joint_groupnorm(x, gamma, beta, groups, eps=1e-6, film=None, silu=False,
                     p_drop=0.0):
     """GroupNorm(+FiLM)(+SiLU)(+dropout) fully fused; the dropout mask is a
     counter-based hash regenerated in backward (no mask tensor)."""
-    seed = int(torch.randint(0, 2 ** 31 - 1, (1,)).item()) if p_drop > 0 else 0
+    seed = _next_gn_seed(x.device) if p_drop > 0 else None
     return _GnFused.apply(x, gamma, beta, film, groups, eps, silu,
                           p_drop, seed)
 

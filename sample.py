@@ -26,8 +26,10 @@ def main() -> None:
     ap.add_argument("--folder", default=None,
                     help="SRN dataset root for conditioning views; synthetic "
                          "conditioning if omitted")
-    ap.add_argument("--model", default="small")
-    ap.add_argument("--sidelength", type=int, default=64)
+    ap.add_argument("--model", default=None,
+                    help="model config name; default: read from checkpoint")
+    ap.add_argument("--sidelength", type=int, default=None,
+                    help="image sidelength; default: read from checkpoint")
     ap.add_argument("--batch-size", type=int, default=1)
     ap.add_argument("--steps", type=int, default=1000)
     ap.add_argument("--guidance", type=float, default=3.0)
@@ -37,7 +39,21 @@ def main() -> None:
     args = ap.parse_args()
 
     device = "cuda" if torch.cuda.is_available() else "cpu"
-    model = XUNet(XUNetConfig.named(args.model), args.sidelength).to(device)
+    # the checkpoint's extra payload records the model config and sidelength;
+    # CLI flags override, and are required only for configless checkpoints
+    payload = torch.load(args.checkpoint, map_location="cpu",
+                         weights_only=False)
+    extra = payload.get("extra", {}) if isinstance(payload, dict) else {}
+    if args.model is not None:
+        model_cfg = XUNetConfig.named(args.model)
+    elif extra.get("model_cfg"):
+        model_cfg = XUNetConfig(**extra["model_cfg"])
+    else:
+        model_cfg = XUNetConfig.named("small")
+    sidelength = (args.sidelength if args.sidelength is not None
+                  else int(extra.get("img_sidelength", 64)))
+    args.sidelength = sidelength
+    model = XUNet(model_cfg, sidelength).to(device)
     ckpt.load_checkpoint(args.checkpoint, model, map_location=device)
     model.eval()
 
