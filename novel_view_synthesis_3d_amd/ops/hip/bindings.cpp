@@ -24,6 +24,8 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias,
                           c10::optional<torch::Tensor> residual,
                           double out_scale);
+std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
+                                         bool with_bias);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v);
 torch::Tensor im2col3x3(torch::Tensor x, int64_t stride, int64_t nplanes,
@@ -59,6 +61,7 @@ TORCH_LIBRARY(nvs3d, m) {
         "Tensor numels, float lr, float b1, float b2, float eps, "
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias, Tensor? residual, float out_scale) -> Tensor");
+  m.def("conv3x3_wgrad(Tensor x, Tensor dy, bool with_bias) -> Tensor[]");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
   m.def("im2col3x3(Tensor x, int stride, int nplanes, int m0, int m1, Tensor? out_buf) -> Tensor");
   m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");
@@ -76,6 +79,7 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("rays_posenc", rays_posenc_py);
   m.impl("fused_adam", fused_adam);
   m.impl("conv3x3_fwd", conv3x3_fwd);
+  m.impl("conv3x3_wgrad", conv3x3_wgrad);
   m.impl("attn_fwd", attn_fwd);
   m.impl("im2col3x3", im2col3x3);
   m.impl("attn_p_from_lse", attn_p_from_lse);

@@ -1,0 +1,335 @@
+// Conv3x3 weight gradient — direct MFMA kernel, CDNA4 gfx950.
+//
+// SURVEY.md §2.4 K1 backward (reference /root/reference/model/xunet.py:81,85
+// under jax.value_and_grad, train.py:70): for the SAME, stride-1, 3x3 frame
+// conv,  dW[co,ky,kx,ci] = sum_{img,y,x} dy[img,y,x,co] * x[img,y+ky-1,x+kx-1,ci].
+//
+// GEMM view per (ky,kx) tap: D[M=co][N=ci] = sum_K A[co][K] B[K][ci] with
+// K = pixels. Both operands are stored pixel-major ([K][C]), so both MFMA
+// fragments need K(pixel)-contiguous elements per lane: the kernel stages
+// dy and x row-tiles into LDS in [csub(16ch)][pixel][16ch-interleave] form
+// and consumes them with ds_read_b64_tr_b16 hardware transpose reads (the
+// attention-V recipe of cdna_hip_programming.md §5.5 T10).
+//
+// Structure:
+//  * One block = (co-tile 128) x (ci-tile 64) x (a contiguous range of
+//    row-tiles). 8 waves as 4(M) x 2(N), each wave one 32x32 output tile
+//    per tap via mfma_f32_32x32x16_bf16 (K-step = 16 pixels).
+//  * ALL NINE (ky,kx) taps accumulate concurrently (9 f32x16 accumulators,
+//    144 regs -> the unified VGPR/AGPR file). The ky dimension pairs the
+//    dy row r with x rows r-1,r,r+1 held in a 4-slot LDS ring; the kx
+//    dimension is a +-1 pixel shift = a +-32 B LDS address offset against
+//    the x row image, which is staged with a 1-pixel halo per row
+//    ([PW+2] pixels; halo = real neighbor pixels of the column tile, or
+//    zero at image edges). So each staged byte of x feeds 9 taps and each
+//    dy byte 9 MFMAs: the kernel reads x and dy from HBM exactly once per
+//    (co,ci)-tile (no 3x/9x im2col materialization).
+//  * Staging is global_load_lds (16 B) with source-permuted addresses
+//    (rule 21): the LDS image is a pure 8-channel-octet permutation of the
+//    global row, so glds stays lane-linear. Subtile stride is padded to
+//    ~= 128 (mod 256) bytes so the 32-lane tr-read service groups land on
+//    disjoint bank halves (conflict-free).
+//  * Split-K: each block atomically adds its fp32 9x32x32-per-wave tile
+//    into dw_acc (Cout,3,3,Cin); bias grad (sum over dy) rides along on
+//    the A fragments of the ci-tile-0 blocks.
+//
+// Constraints: Cin % 64 == 0, Cout % 128 == 0, W % 16 == 0 and
+// (W <= 128 or W % 128 == 0). Dispatch falls back to the im2col+GEMM path
+// otherwise (ops/hip_ops.py).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+using as3_bf16x4p = __attribute__((address_space(3))) bf16x4*;
+
+using as1_cvp = const __attribute__((address_space(1))) void*;
+using as3_vp = __attribute__((address_space(3))) void*;
+__device__ __forceinline__ as1_cvp as_global(const void* p) {
+  return (as1_cvp)(unsigned long long)(uintptr_t)p;
+}
+__device__ __forceinline__ as3_vp as_shared(void* p) {
+  return (as3_vp)(unsigned int)(uintptr_t)p;
+}
+
+__device__ __forceinline__ bf16x8 tr16x8(const char* p) {
+  bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((as3_bf16x4p)(const_cast<char*>(p)));
+  bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((as3_bf16x4p)(const_cast<char*>(p) + 128));
+  return __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
+constexpr int WBM = 128;   // co per block
+constexpr int WBN = 64;    // ci per block
+constexpr int WT = 512;    // 8 waves
+
+struct WgShape {
+  int IMG, H, W, Cin, Cout;
+  int PW;          // column-tile width (min(W,128)), multiple of 16
+  int nct;         // column tiles per row = W / PW
+  int sd;          // dy image: bytes per 16-ch subtile (PW*32 + pad)
+  int sx;          // x image: bytes per 16-ch subtile ((PW+2)*32 + pad)
+  int units;       // IMG*H*nct row-tiles
+  int nb_m, nb_n;  // Cout/WBM, Cin/WBN
+  int sk;          // split-K factor
+  int with_bias;
+};
+
+// stride pad so that stride % 256 == 128 (disjoint bank halves for the
+// two 16-lane tr-read groups of each 32-lane service group)
+static int pad128mod256(int base) {
+  return base + ((128 - (base % 256)) + 256) % 256;
+}
+
+__global__ __launch_bounds__(WT)
+void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
+                          const bf16* __restrict__ dy,
+                          const bf16* __restrict__ zbuf,
+                          float* __restrict__ dwacc,  // (Cout,3,3,Cin) zeroed
+                          float* __restrict__ dbacc,  // (Cout,) zeroed or null
+                          WgShape s) {
+  // block decode: bid = sk * (nb_m*nb_n) + (bm*nb_n + bn), with the
+  // bijective XCD remap so blocks sharing a split-K data range (identical
+  // dy/x reads, different (co,ci) tiles) sit on one XCD's L2.
+  int bid = blockIdx.x;
+  const int nwg = s.sk * s.nb_m * s.nb_n;
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int mn = bid % (s.nb_m * s.nb_n);
+  const int sk = bid / (s.nb_m * s.nb_n);
+  const int bm = mn / s.nb_n;
+  const int bn = mn % s.nb_n;
+  const int co0 = bm * WBM;
+  const int ci0 = bn * WBN;
+
+  // this block's contiguous row-tile range [u0, u1), ct-major so that
+  // consecutive units usually share the x-row ring
+  const int per = (s.units + s.sk - 1) / s.sk;
+  const int u0 = sk * per;
+  const int u1 = min(s.units, u0 + per);
+  if (u0 >= u1) return;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: [dy buf 0][dy buf 1][x slot 0..3]  (all offsets arithmetic —
+  // no runtime-indexed pointer arrays, §5.4 rule 20)
+  const int dy_bytes = 8 * s.sd;        // 8 subtiles (128 co)
+  const int x_bytes = 4 * s.sx;         // 4 subtiles (64 ci)
+  char* xbase = smem + 2 * dy_bytes;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 1;   // 0..3: co quarter (32 rows)
+  const int wn = wave & 1;    // 0..1: ci half (32 cols)
+
+  // per-lane tr-read base offsets (within an operand image)
+  const int g = lane >> 4;       // 16-lane transpose group
+  const int m16 = lane & 15;
+  const int lane_px = ((g >> 1) * 8 + (m16 >> 2)) * 32 + (m16 & 3) * 8;
+  const int a_off = (wm * 2 + (g & 1)) * s.sd + lane_px;
+  const int b_off = (wn * 2 + (g & 1)) * s.sx + lane_px;
+
+  // ---- staging (glds, source-permuted) -------------------------------
+  // dy row-tile -> dybuf[buf]; x row-tile -> x slot. Each wave covers LDS
+  // bytes [c*1024 + lane*16) for chunks c = wave, wave+8, ...
+  const long rowstride_x = (long)s.W * s.Cin;
+  const long rowstride_dy = (long)s.W * s.Cout;
+
+  auto stage_dy = [&](int img, int r, int ct, int buf) {
+    const long base = ((long)img * s.H + r) * rowstride_dy
+                      + (long)ct * s.PW * s.Cout + co0;
+    char* dst = smem + buf * dy_bytes;
+    const int tot = dy_bytes;
+    for (int o = wave * 1024 + lane * 16; o < tot; o += 8 * 1024) {
+      const int csub = o / s.sd;
+      const int w = o % s.sd;
+      const bf16* src = zbuf;
+      if (w < s.PW * 32) {
+        const int px = w >> 5;
+        const int oct = (w & 31) >> 4;
+        src = dy + base + (long)px * s.Cout + csub * 16 + oct * 8;
+      }
+      __builtin_amdgcn_global_load_lds(as_global(src),
+          as_shared(dst + (o - lane * 16)), 16, 0, 0);
+    }
+  };
+
+  auto stage_x = [&](int img, int r, int ct, int slot) {
+    const long base = ((long)img * s.H + r) * rowstride_x + ci0;
+    const int px0 = ct * s.PW - 1;  // image pixel of halo index 0
+    char* dst = xbase + slot * x_bytes;
+    const int tot = x_bytes;
+    for (int o = wave * 1024 + lane * 16; o < tot; o += 8 * 1024) {
+      const int csub = o / s.sx;
+      const int w = o % s.sx;
+      const bf16* src = zbuf;
+      if (w < (s.PW + 2) * 32) {
+        const int px = px0 + (w >> 5);
+        const int oct = (w & 31) >> 4;
+        if (px >= 0 && px < s.W) {
+          src = x + base + (long)px * s.Cin + csub * 16 + oct * 8;
+        }
+      }
+      __builtin_amdgcn_global_load_lds(as_global(src),
+          as_shared(dst + (o - lane * 16)), 16, 0, 0);
+    }
+  };
+
+  // ---- accumulators ---------------------------------------------------
+  float acc[9][16];
+#pragma unroll
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[t][e] = 0.f;
+  float dbsum = 0.f;
+  const bool do_bias = s.with_bias && bn == 0 && wn == 0;
+
+  // ---- compute one staged row-tile ------------------------------------
+  // dy row r in dybuf[buf]; x rows r-1..r+1 in ring slots (rr+1)&3; the
+  // valid mask enables the ky taps whose x row exists.
+  auto compute_row = [&](int buf, int r, int vmask) {
+    const char* abase = smem + buf * dy_bytes + a_off;
+#pragma unroll 1
+    for (int p0 = 0; p0 < s.PW; p0 += 16) {
+      bf16x8 af = tr16x8(abase + p0 * 32);
+      if (do_bias) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) dbsum += (float)af[e];
+      }
+#pragma unroll
+      for (int ky = 0; ky < 3; ++ky) {
+        if (!(vmask & (1 << ky))) continue;
+        const int slot = (r + ky) & 3;   // x row r+ky-1 -> slot (r+ky-1+1)&3
+        const char* bb = xbase + slot * x_bytes + b_off + p0 * 32;
+#pragma unroll
+        for (int kx = 0; kx < 3; ++kx) {
+          bf16x8 bf = tr16x8(bb + kx * 32);
+          *reinterpret_cast<f32x16*>(acc[ky * 3 + kx]) =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  af, bf, *reinterpret_cast<f32x16*>(acc[ky * 3 + kx]),
+                  0, 0, 0);
+        }
+      }
+    }
+  };
+
+  // ---- main loop over row-tiles ---------------------------------------
+  // Segment = maximal run of consecutive rows of one (img, ct): prologue
+  // stages x rows r0-1..r0+1 and dy r0, then each iteration prefetches the
+  // next row's dy and x(r+2) while computing row r.
+  int u = u0;
+  while (u < u1) {
+    const int ct = u / (s.IMG * s.H);
+    const int vr = u % (s.IMG * s.H);
+    const int img = vr / s.H;
+    const int r0 = vr % s.H;
+    // segment = run of rows of this (img, ct); img/ct boundaries end it
+    const int rend = min(s.H, r0 + (u1 - u));
+
+    // prologue staging for row r0
+    if (r0 > 0) stage_x(img, r0 - 1, ct, r0 & 3);
+    stage_x(img, r0, ct, (r0 + 1) & 3);
+    if (r0 + 1 < s.H) stage_x(img, r0 + 1, ct, (r0 + 2) & 3);
+    stage_dy(img, r0, ct, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int r = r0; r < rend; ++r) {
+      const int buf = (r - r0) & 1;
+      // prefetch next row's inputs while computing this one
+      if (r + 1 < rend) {
+        stage_dy(img, r + 1, ct, buf ^ 1);
+        if (r + 2 < s.H) stage_x(img, r + 2, ct, (r + 3) & 3);
+      }
+      const int vmask = (r > 0 ? 1 : 0) | 2 | (r + 1 < s.H ? 4 : 0);
+      compute_row(buf, r, vmask);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
+    u += rend - r0;
+  }
+
+  // ---- epilogue: fp32 atomic reduction into dwacc ---------------------
+  // D layout (mfma_f32_32x32x16): col = lane&31 (ci), row = (e&3) + 8*(e>>2)
+  // + 4*(lane>>5) (co), e in [0,16).
+  const int ci = ci0 + wn * 32 + (lane & 31);
+  const int co_base = co0 + wm * 32 + 4 * (lane >> 5);
+#pragma unroll
+  for (int ky = 0; ky < 3; ++ky)
+#pragma unroll
+    for (int kx = 0; kx < 3; ++kx)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int co = co_base + (e & 3) + 8 * (e >> 2);
+        atomicAdd(dwacc + (((long)co * 3 + ky) * 3 + kx) * s.Cin + ci,
+                  acc[ky * 3 + kx][e]);
+      }
+  if (do_bias) {
+    // lanes l and l+32 hold the same co (different pixels)
+    atomicAdd(dbacc + co0 + wm * 32 + (lane & 31), dbsum);
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
+                                         bool with_bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              dy.scalar_type() == torch::kBFloat16);
+  const int nd = x.dim();
+  TORCH_CHECK(nd == 4 || nd == 5);
+  WgShape s;
+  auto xs = x.sizes();
+  s.IMG = nd == 5 ? xs[0] * xs[1] : xs[0];
+  s.H = xs[nd - 3]; s.W = xs[nd - 2]; s.Cin = xs[nd - 1];
+  s.Cout = dy.size(nd - 1);
+  TORCH_CHECK(dy.numel() == (long)s.IMG * s.H * s.W * s.Cout);
+  TORCH_CHECK(s.Cin % WBN == 0 && s.Cout % WBM == 0);
+  TORCH_CHECK(s.W % 16 == 0 && (s.W <= 128 || s.W % 128 == 0));
+
+  s.PW = std::min(s.W, 128);
+  s.nct = s.W / s.PW;
+  s.sd = pad128mod256(s.PW * 32);
+  s.sx = pad128mod256((s.PW + 2) * 32);
+  s.units = s.IMG * s.H * s.nct;
+  s.nb_m = s.Cout / WBM;
+  s.nb_n = s.Cin / WBN;
+  s.with_bias = with_bias ? 1 : 0;
+  // split-K sized for ~2 block-waves over 256 CUs (1 block/CU at this LDS)
+  const int target = 512;
+  s.sk = std::max(1, std::min(s.units,
+                              target / std::max(1, s.nb_m * s.nb_n)));
+
+  auto opts = x.options().dtype(torch::kFloat);
+  auto dw = torch::zeros({s.Cout, 3, 3, s.Cin}, opts);
+  torch::Tensor db;
+  if (with_bias) db = torch::zeros({s.Cout}, opts);
+
+  static torch::Tensor zbuf;
+  if (!zbuf.defined() || zbuf.device() != x.device()) {
+    zbuf = torch::zeros({64}, x.options());
+  }
+
+  const size_t lds = 2 * (8 * s.sd) + 4 * (4 * s.sx);
+  TORCH_CHECK(lds <= 160 * 1024, "wgrad LDS overflow: ", lds);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = s.sk * s.nb_m * s.nb_n;
+  hipLaunchKernelGGL(conv3x3_wgrad_kernel, dim3(grid), dim3(WT), lds, stream,
+      reinterpret_cast<const bf16*>(x.data_ptr()),
+      reinterpret_cast<const bf16*>(dy.data_ptr()),
+      reinterpret_cast<const bf16*>(zbuf.data_ptr()),
+      dw.data_ptr<float>(),
+      with_bias ? db.data_ptr<float>() : nullptr, s);
+  if (with_bias) return {dw, db};
+  return {dw};
+}

@@ -198,7 +198,7 @@ class _FrameConv3x3(torch.autograd.Function):
             else:
                 dx = _miopen_conv(dy, wt, None)
         if need_w or need_b:
-            dw4, db = _wgrad_im2col_gemm(x, w, dy, ctx.has_bias)
+            dw4, db = _wgrad(x, w, dy, ctx.has_bias)
             if need_w:
                 dw = dw4
         dres = dy if ctx.has_res else None
@@ -219,6 +219,30 @@ def _miopen_conv(x, w, bias):
     y = Fn.conv2d(_nchw_view(x), w.permute(0, 3, 1, 2), bias, padding=1)
     out = y.permute(0, 2, 3, 1)
     return out.reshape(*shape[:-1], w.shape[0]).contiguous()
+
+
+def _wgrad_shapes_supported(x, cout) -> bool:
+    """The direct MFMA wgrad kernel covers W%16==0 and (W<=128 or W%128==0),
+    Cin%64==0, Cout%128==0 — every conv the MFMA fwd kernel takes."""
+    W = x.shape[-2]
+    cin = x.shape[-1]
+    return (cin % 64 == 0 and cout % 128 == 0 and W % 16 == 0
+            and (W <= 128 or W % 128 == 0)
+            and os.environ.get("NVS3D_WGRAD", "mfma") != "im2col")
+
+
+def _wgrad(x, w, dy, has_bias):
+    """Conv wgrad dispatch: hand-written direct MFMA kernel (one pass over
+    x/dy, all 9 taps concurrent, fp32 split-K atomics — conv3x3_wgrad.hip),
+    falling back to the 3-row-shift im2col+hipBLASLt decomposition for
+    unsupported shapes."""
+    cout = w.shape[0]
+    if _wgrad_shapes_supported(x, cout):
+        outs = _OPS.conv3x3_wgrad(x, dy, has_bias)
+        dw = outs[0].to(x.dtype)
+        db = outs[1].to(dy.dtype) if has_bias else None
+        return dw, db
+    return _wgrad_im2col_gemm(x, w, dy, has_bias)
 
 
 def _wgrad_im2col_gemm(x, w, dy, has_bias):

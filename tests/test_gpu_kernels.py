@@ -214,6 +214,43 @@ def test_conv3x3_autograd_parity():
         assert err / scale < 5e-2, f"{n}: rel {err/scale:.3e}"
 
 
+WGRAD_SHAPES = [
+    (1, 2, 16, 16, 256, 128),   # W=16: halo + image-edge rows every 16 px
+    (2, 2, 16, 16, 1024, 128),  # deep level channels, multi-image split
+    (1, 2, 32, 32, 64, 128),    # W=32
+    (1, 2, 64, 64, 128, 256),   # L1-ish
+    (2, 2, 128, 16, 64, 128),   # W=128 (full-config L0 width), multi-image
+]
+
+
+@pytest.mark.parametrize("shape", WGRAD_SHAPES)
+def test_conv3x3_wgrad_kernel_parity(shape):
+    """Direct MFMA wgrad kernel (conv3x3_wgrad.hip) vs fp32 autograd oracle.
+
+    Matches the bwd of /root/reference/model/xunet.py:81,85 under
+    train.py:70. H and W are passed independently to exercise row-ring
+    boundaries (H small, W large)."""
+    B, F, H, W, Cin, Cout = shape
+    g = torch.Generator(device="cuda").manual_seed(3)
+    x = torch.randn(B, F, H, W, Cin, device="cuda", generator=g,
+                    dtype=torch.bfloat16)
+    w = torch.zeros(Cout, 3, 3, Cin, device="cuda")
+    dy = torch.randn(B, F, H, W, Cout, device="cuda", generator=g,
+                     dtype=torch.bfloat16)
+    outs = torch.ops.nvs3d.conv3x3_wgrad(x, dy, True)
+    dw, db = outs[0], outs[1]
+
+    xr = x.float().detach().requires_grad_(True)
+    wr = w.detach().requires_grad_(True)
+    br = torch.zeros(Cout, device="cuda", requires_grad=True)
+    y = ref.frame_conv3x3(xr, wr, br)
+    y.backward(dy.float())
+    for name, got, want in (("dw", dw, wr.grad), ("db", db, br.grad)):
+        err = (got - want).abs().max().item()
+        scale = want.abs().max().item() + 1e-6
+        assert err / scale < 3e-2, f"{name}: rel {err/scale:.3e}"
+
+
 ATTN_SHAPES = [
     (2, 1024, 4, 128),   # full config res 32
     (2, 256, 4, 256),    # full config res 16
