@@ -403,6 +403,15 @@ class _Attention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         B, L, H, D = q.shape
+        Lk = k.shape[1]
+        if (D in (16, 32, 64, 128) and L % 128 == 0 and Lk % 128 == 0
+                and os.environ.get("NVS3D_ATTN_BWD", "fused") != "gemm"):
+            # fused flash-style backward (attn_bwd.hip): tile-wise recompute
+            # of S/P from q/k/lse — no (B,H,L,L) tensor touches HBM
+            doc = do.to(q.dtype).contiguous()
+            delta = _OPS.attn_delta(doc, o)
+            dq, dk, dv = _OPS.attn_bwd_fused(q, k, v, doc, lse, delta)
+            return dq, dk, dv
         scale = 1.0 / (D ** 0.5)
         # (B,H,L,D) strided views; rocBLAS consumes them without copies
         qt = q.permute(0, 2, 1, 3)
