@@ -183,9 +183,10 @@ class Trainer:
         if raw is None:
             raw = self.next_batch()
         batch, cond_mask, noise = self.prepare_model_inputs(raw)
-        self.ddp.zero_flags()
-        # keep grad storage stable on GPU so FusedAdam's pointer plan caches
-        self.opt.zero_grad(set_to_none=(self.device.type != "cuda"))
+        self.ddp.zero_flags()  # zeroes the flat grad buffers when DP is on
+        if not self.ddp.enabled:
+            # keep grad storage stable on GPU so FusedAdam's plan caches
+            self.opt.zero_grad(set_to_none=(self.device.type != "cuda"))
         with self._autocast():
             out = self.model(batch, cond_mask)
         loss = self.compute_loss(out, noise)
@@ -272,28 +273,9 @@ class Trainer:
             for k, v in raw.items():
                 self._static_raw[k].copy_(v, non_blocking=True)
         self._graph.replay()
-        if self.ddp.enabled:
-            self._allreduce_grads_post()
+        self.ddp.reduce_all()  # in-place on the flat grad buffers
         self.opt.step()
         return self._static_loss.detach()
-
-    def _allreduce_grads_post(self):
-        """Bucketed all-reduce of the grads a replay produced (the in-graph
-        backward cannot launch RCCL ops)."""
-        for b in self.ddp.buckets:
-            b.ready = len(b.params)
-            b.launched = False
-        for b in self.ddp.buckets:
-            self.ddp._launch(b)
-        inv = 1.0 / self.ddp.world
-        for b in self.ddp.buckets:
-            b.handle.wait()
-            b.buffer.mul_(inv)
-            off = 0
-            for p in b.params:
-                n = p.numel()
-                p.grad.reshape(-1).copy_(b.buffer[off:off + n])
-                off += n
 
     def train(self) -> None:
         """Run to train_num_steps. Interrupts (SIGINT/SIGTERM) and crashes

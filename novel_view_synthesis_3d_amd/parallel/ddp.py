@@ -65,14 +65,18 @@ class _Bucket:
 
 
 class DataParallelEngine:
-    """Bucketed gradient all-reduce, overlapped with backward.
+    """Bucketed gradient all-reduce with PERSISTENT FLAT GRAD STORAGE:
+    every param's .grad is a view into its bucket's contiguous fp32 buffer
+    (what torch DDP calls gradient_as_bucket_view), so the all-reduce runs
+    in place on the gradients themselves — zero staging copies per step.
+    With RCCL the reduction uses ReduceOp.AVG (no separate 1/N pass).
 
     Usage:
         engine = DataParallelEngine(model, bucket_mb=40)
         for step ...:
-            engine.zero_flags()
-            loss.backward()          # hooks launch async all-reduces
-            engine.finish()          # wait + average
+            engine.zero_flags()      # zeroes the flat buffers (= the grads)
+            loss.backward()          # hooks launch async in-place reduces
+            engine.finish()          # wait (+ 1/N on gloo)
             optimizer.step()
     """
 
@@ -88,6 +92,8 @@ class DataParallelEngine:
         self._param_bucket = {}
         if not self.enabled:
             return
+        # ReduceOp.AVG is an RCCL-native fused sum+scale; gloo needs SUM+mul
+        self._avg = dist.get_backend(self.group) == "nccl"
 
         # Identical start state on every rank (reference defect D3 fix).
         with torch.no_grad():
@@ -109,6 +115,18 @@ class DataParallelEngine:
         if bucket.numel:
             self.buckets.append(bucket)
 
+        # flat buffers; each grad becomes a view (autograd accumulates into
+        # existing .grad in place, so the views persist across steps)
+        for b in self.buckets:
+            b.buffer = torch.zeros(b.numel, dtype=torch.float32,
+                                   device=b.params[0].device)
+            off = 0
+            for p in b.params:
+                n = p.numel()
+                p.grad = b.buffer[off:off + n].view(p.shape)
+                off += n
+        self._buffers = [b.buffer for b in self.buckets]
+
         for p in params:
             h = p.register_post_accumulate_grad_hook(self._on_grad)
             self._hooks.append(h)
@@ -123,47 +141,50 @@ class DataParallelEngine:
             self._launch(b)
 
     def _launch(self, b: _Bucket) -> None:
-        if b.buffer is None or b.buffer.device != b.params[0].grad.device:
-            b.buffer = torch.empty(
-                b.numel, dtype=torch.float32, device=b.params[0].grad.device)
-        off = 0
-        for p in b.params:
-            n = p.numel()
-            b.buffer[off:off + n].copy_(p.grad.reshape(-1))
-            off += n
-        b.handle = dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
-                                   group=self.group, async_op=True)
+        op = dist.ReduceOp.AVG if self._avg else dist.ReduceOp.SUM
+        b.handle = dist.all_reduce(b.buffer, op=op, group=self.group,
+                                   async_op=True)
         b.launched = True
 
     def zero_flags(self) -> None:
+        """Reset per-step state AND zero the flat grad buffers (replaces
+        optimizer.zero_grad — .grad views must never be detached)."""
         for b in self.buckets:
             b.ready = 0
             b.handle = None
             b.launched = False
+        if self.enabled:
+            torch._foreach_zero_(self._buffers)
 
     def finish(self) -> None:
-        """Wait for all reductions; write averaged grads back."""
+        """Wait for all in-place reductions (launch any bucket whose params
+        produced no grad hook this step — its buffer holds zeros/partials)."""
         if not (self.enabled and self._sync):
             return
-        inv = 1.0 / self.world
         for b in self.buckets:
             if not b.launched:
-                # params without grads this step (shouldn't happen in X-UNet,
-                # but stay correct): reduce whatever accumulated
-                if any(p.grad is not None for p in b.params):
-                    for p in b.params:
-                        if p.grad is None:
-                            p.grad = torch.zeros_like(p)
-                    self._launch(b)
-                else:
-                    continue
+                self._launch(b)
+        inv = 1.0 / self.world
+        for b in self.buckets:
             b.handle.wait()
-            b.buffer.mul_(inv)
-            off = 0
-            for p in b.params:
-                n = p.numel()
-                p.grad.reshape(-1).copy_(b.buffer[off:off + n])
-                off += n
+            if not self._avg:
+                b.buffer.mul_(inv)
+
+    def reduce_all(self) -> None:
+        """Synchronous whole-model reduce (used after a hipGraph replay,
+        where the in-graph backward cannot launch RCCL ops)."""
+        if not self.enabled:
+            return
+        for b in self.buckets:
+            b.ready = len(b.params)
+            b.launched = False
+        for b in self.buckets:
+            self._launch(b)
+        inv = 1.0 / self.world
+        for b in self.buckets:
+            b.handle.wait()
+            if not self._avg:
+                b.buffer.mul_(inv)
 
     @contextmanager
     def no_sync(self):

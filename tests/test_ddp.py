@@ -109,3 +109,73 @@ def test_ddp_gradient_equivalence(tmp_path):
     # average of shard losses == full-batch loss for equal shards (MSE mean)
     assert torch.allclose(g0, ref, atol=1e-5), \
         (g0 - ref).abs().max().item()
+
+
+def _worker4(rank, world, port, results_dir):
+    """World-4 variant: two steps (flat-grad views must survive re-zeroing)
+    + the reduce_all path (what a hipGraph replay uses)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from novel_view_synthesis_3d_amd.parallel.ddp import DataParallelEngine
+
+        model = _build_model(seed=200 + rank)
+        engine = DataParallelEngine(model, bucket_mb=0.25)
+        batch, noise = _full_batch(B=4, H=16)
+        sl = slice(rank, rank + 1)
+        shard = {k: v[sl] for k, v in batch.items()}
+
+        # step 1: hook-launched overlapped path
+        engine.zero_flags()
+        out = model(shard, cond_mask=torch.ones(1))
+        torch.nn.functional.mse_loss(out, noise[sl]).backward()
+        engine.finish()
+        g_hook = torch.cat([p.grad.reshape(-1)
+                            for p in model.parameters()]).clone()
+        # grads must literally BE bucket-buffer views (no staging copies)
+        for b in engine.buckets:
+            assert b.params[0].grad.data_ptr() == b.buffer.data_ptr()
+
+        # step 2: same shard through the reduce_all (graph-replay) path;
+        # the re-zeroed views must produce identical averaged grads
+        engine.zero_flags()
+        with engine.no_sync():
+            out = model(shard, cond_mask=torch.ones(1))
+            torch.nn.functional.mse_loss(out, noise[sl]).backward()
+        engine.reduce_all()
+        g_ra = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+        assert torch.allclose(g_hook, g_ra, atol=1e-6), \
+            (g_hook - g_ra).abs().max().item()
+        torch.save({"grads": g_hook},
+                   os.path.join(results_dir, f"w4_rank{rank}.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_world4_flat_views_and_reduce_all(tmp_path):
+    world = 4
+    port = 29523
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker4, args=(r, world, port, str(tmp_path)))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+
+    model = _build_model(seed=200)
+    batch, noise = _full_batch(B=4, H=16)
+    out = model(batch, cond_mask=torch.ones(4))
+    torch.nn.functional.mse_loss(out, noise).backward()
+    ref = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+    for r in range(world):
+        g = torch.load(tmp_path / f"w4_rank{r}.pt",
+                       weights_only=False)["grads"]
+        assert torch.allclose(g, ref, atol=1e-5), \
+            (g - ref).abs().max().item()
