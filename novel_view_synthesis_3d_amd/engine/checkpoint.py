@@ -61,18 +61,110 @@ def export_flax_npz(model: torch.nn.Module, path: str) -> None:
     np.savez(path, **arrays)
 
 
-def import_flax_npz(model: torch.nn.Module, path: str, strict: bool = True) -> None:
-    data = np.load(path)
+def _load_flax_arrays(model, arrays: Dict[str, np.ndarray],
+                      strict: bool) -> None:
+    """Copy a {flax_path: ndarray} dict into the model (layout transforms
+    from flax_tree). Handles two reference quirks:
+      * pmap-stacked checkpoints (leading device axis): takes
+        replica 0 when the array has exactly one extra leading dim;
+      * bfloat16 arrays arrive as uint16 bit patterns (numpy has no bf16).
+    """
     tree = flax_tree(model)
-    missing = set(tree) - set(data.files)
-    extra = set(data.files) - set(tree)
+    missing = set(tree) - set(arrays)
+    extra = set(arrays) - set(tree)
     if strict and (missing or extra):
         raise KeyError(f"flax tree mismatch: missing={sorted(missing)[:5]} "
                        f"extra={sorted(extra)[:5]}")
     with torch.no_grad():
-        for k, (p, _, from_f) in tree.items():
-            if k in data:
-                p.copy_(from_f(torch.from_numpy(data[k])).to(p.dtype))
+        for k, (p, to_f, from_f) in tree.items():
+            if k not in arrays:
+                continue
+            a = arrays[k]
+            want = tuple(to_f(p.detach().cpu()).shape)
+            if a.ndim == len(want) + 1 and tuple(a.shape[1:]) == want:
+                a = a[0]  # pmap device axis (reference train.py:161-167)
+            if a.dtype == np.uint16:  # bf16 bit pattern
+                t = torch.from_numpy(np.ascontiguousarray(a)).view(
+                    torch.bfloat16).float()
+            else:
+                t = torch.from_numpy(np.ascontiguousarray(a))
+            p.copy_(from_f(t).to(p.dtype))
+
+
+def import_flax_npz(model: torch.nn.Module, path: str, strict: bool = True) -> None:
+    data = np.load(path)
+    _load_flax_arrays(model, {k: data[k] for k in data.files}, strict)
+
+
+# ---------------------------------------------------------------------------
+# flax msgpack interchange — the reference's actual checkpoint format
+# (flax.training.checkpoints.save_checkpoint writes
+# flax.serialization.to_bytes(params) to `<dir>/<prefix><step>`:
+# /root/reference/train.py:161-167, sampling.py:106-114). Wire format:
+# msgpack maps of maps; ndarray leaves are ExtType 1 wrapping
+# packb((shape, dtype.name, raw C-order bytes)) — flax/serialization.py.
+# ---------------------------------------------------------------------------
+
+_EXT_NDARRAY = 1
+
+
+def _flax_ext_pack(x):
+    import msgpack
+    if isinstance(x, np.ndarray):
+        payload = msgpack.packb((x.shape, x.dtype.name, x.tobytes("C")),
+                                use_bin_type=True)
+        return msgpack.ExtType(_EXT_NDARRAY, payload)
+    raise TypeError(f"cannot pack {type(x)}")
+
+
+def _flax_ext_unpack(code, data):
+    import msgpack
+    if code != _EXT_NDARRAY:
+        return msgpack.ExtType(code, data)
+    shape, dtype_name, buf = msgpack.unpackb(data, raw=True)
+    if isinstance(dtype_name, bytes):
+        dtype_name = dtype_name.decode()
+    if dtype_name == "bfloat16":  # jax ml_dtypes name; keep the bits
+        return np.frombuffer(buf, dtype=np.uint16).reshape(shape)
+    return np.frombuffer(buf, dtype=np.dtype(dtype_name)).reshape(shape)
+
+
+def _flatten(tree, prefix="", out=None):
+    if out is None:
+        out = {}
+    for k, v in tree.items():
+        key = k.decode() if isinstance(k, bytes) else str(k)
+        if isinstance(v, dict):
+            _flatten(v, prefix + key + "/", out)
+        else:
+            out[prefix + key] = v
+    return out
+
+
+def export_flax_msgpack(model: torch.nn.Module, path: str) -> None:
+    """Write the param tree in the reference's flax msgpack format."""
+    import msgpack
+    nested: Dict = {}
+    for k, (p, to_f, _) in flax_tree(model).items():
+        node = nested
+        parts = k.split("/")
+        for part in parts[:-1]:
+            node = node.setdefault(part, {})
+        node[parts[-1]] = to_f(p.detach().cpu()).numpy()
+    with open(path, "wb") as f:
+        f.write(msgpack.packb(nested, default=_flax_ext_pack,
+                              strict_types=True))
+
+
+def import_flax_msgpack(model: torch.nn.Module, path: str,
+                        strict: bool = True) -> None:
+    """Load a flax msgpack checkpoint (as written by the reference trainer)
+    into the model, transposing flax layouts and un-stacking pmap replicas."""
+    import msgpack
+    with open(path, "rb") as f:
+        tree = msgpack.unpackb(f.read(), ext_hook=_flax_ext_unpack,
+                               raw=False, strict_map_key=False)
+    _load_flax_arrays(model, _flatten(tree), strict)
 
 
 # ---------------------------------------------------------------------------

@@ -521,12 +521,13 @@ def test_gn_fused_dropout():
                     dtype=torch.bfloat16)
     gm = torch.ones(C, device="cuda")
     bt = torch.zeros(C, device="cuda")
+    seed = torch.full((1,), 12345, dtype=torch.int32, device="cuda")
     y0, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
-                                      0.0, 0)
+                                      0.0, None)
     y1, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
-                                      p, 12345)
+                                      p, seed)
     y2, _, _ = torch.ops.nvs3d.gn_fwd(x, gm, bt, None, 32, 1e-6, True,
-                                      p, 12345)
+                                      p, seed)
     # same seed -> same MASK (values can differ in the last ulp: the GN
     # stats reduction uses LDS float atomics, so summation order varies)
     assert torch.equal(y1 == 0, y2 == 0)

@@ -1,0 +1,71 @@
+#!/usr/bin/env python3
+"""Bisect the graph-replay NaN at full config: run the captured training
+step under several ablations and report per-replay losses.
+
+Run on a GPU box:  python tools/debug_graph.py
+"""
+
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, ".")
+
+
+def run_variant(name, dropout, attn, env=None, steps=6, batch=4):
+    import importlib
+
+    env = env or {}
+    old = {}
+    for k, v in env.items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
+    try:
+        from novel_view_synthesis_3d_amd.config import (TrainConfig,
+                                                        XUNetConfig)
+        from novel_view_synthesis_3d_amd.engine.trainer import Trainer
+
+        mcfg = XUNetConfig.full()
+        mcfg.dropout = dropout
+        if not attn:
+            mcfg.attn_resolutions = ()
+        tcfg = TrainConfig()
+        tcfg.use_graph = True
+        tcfg.amp = "bf16"
+        tcfg.data = "synthetic"
+        losses = []
+        tr = Trainer(folder=None, train_batch_size=batch,
+                     train_num_steps=steps, img_sidelength=128,
+                     model_cfg=mcfg, train_cfg=tcfg)
+        for i in range(steps):
+            loss = tr.train_step()
+            torch.cuda.synchronize()
+            losses.append(float(loss.item()))
+        print(f"[{name}] losses: "
+              + " ".join(f"{v:.4f}" for v in losses), flush=True)
+        del tr
+        torch.cuda.empty_cache()
+    except Exception as e:
+        print(f"[{name}] FAILED: {e!r}", flush=True)
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+
+
+def main():
+    torch.manual_seed(0)
+    run_variant("baseline-graph", dropout=0.1, attn=True)
+    run_variant("no-dropout", dropout=0.0, attn=True)
+    run_variant("no-attn", dropout=0.1, attn=False)
+    run_variant("attn-bwd-gemm", dropout=0.1, attn=True,
+                env={"NVS3D_ATTN_BWD": "gemm"})
+    run_variant("wgrad-im2col", dropout=0.1, attn=True,
+                env={"NVS3D_WGRAD": "im2col"})
+
+
+if __name__ == "__main__":
+    main()
