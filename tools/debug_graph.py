@@ -58,6 +58,16 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4):
 
 def main():
     torch.manual_seed(0)
+    import sys
+    if len(sys.argv) > 1 and sys.argv[1] == "batch":
+        run_variant("graph-b8", dropout=0.1, attn=True, batch=8)
+        run_variant("graph-b12", dropout=0.1, attn=True, batch=12)
+        run_variant("graph-b16", dropout=0.1, attn=True, batch=16)
+        run_variant("graph-b16-nodrop", dropout=0.0, attn=True, batch=16)
+        run_variant("graph-b16-libpaths", dropout=0.1, attn=True, batch=16,
+                    env={"NVS3D_ATTN_BWD": "gemm", "NVS3D_WGRAD": "im2col"})
+        run_variant("graph-b16-noattn", dropout=0.1, attn=False, batch=16)
+        return
     run_variant("baseline-graph", dropout=0.1, attn=True)
     run_variant("no-dropout", dropout=0.0, attn=True)
     run_variant("no-attn", dropout=0.1, attn=False)
