@@ -89,14 +89,32 @@ def joint_groupnorm(x, gamma, beta, groups: int, eps: float = 1e-6,
                                p_drop)
 
 
-def attention(q, k, v):
+def attention(q, k, v, kv_swap: bool = False):
+    """kv_swap=True: batch b's queries attend batch b^1's keys/values —
+    the model's cross-FRAME attention with both frames batched as (B*2)."""
     if _use_hip(q, "attention"):
         B, L, h, d = q.shape
         bf16_path = (q.dtype == torch.bfloat16
                      or torch.is_autocast_enabled())
         if bf16_path and L % 64 == 0 and d in (16, 32, 64, 128, 256):
-            return _HIP_MOD.attention(q, k, v)
+            return _HIP_MOD.attention(q, k, v, kv_swap)
+    if kv_swap:
+        B2 = k.shape[0]
+        k = k.reshape(B2 // 2, 2, *k.shape[1:]).flip(1).reshape(k.shape)
+        v = v.reshape(B2 // 2, 2, *v.shape[1:]).flip(1).reshape(v.shape)
     return ref.attention(q, k, v)
+
+
+def linear(x, w, b=None):
+    """nn.Linear forward; on MI355X the backward uses the split-K MFMA
+    wgrad kernel for the tall-skinny shapes hipBLASLt collapses on."""
+    if _use_hip(x, "linear"):
+        bf16_path = (x.dtype == torch.bfloat16
+                     or torch.is_autocast_enabled())
+        if bf16_path:
+            return _HIP_MOD.linear(x, w, b)
+    import torch.nn.functional as F
+    return F.linear(x, w, b)
 
 
 def nearest_upsample2x(x):

@@ -41,6 +41,8 @@ struct ABwdShape {
   long sqb, sqt;   // q strides (elements); head stride = D, d contiguous
   long skb, skt;   // k
   long svb, svt;   // v
+  int kv_swap;     // batch b's queries attend batch b^1's k/v (batched
+                   // cross-frame attention; see attn_fwd.hip)
 };
 
 __device__ __forceinline__ int swz_row(int row, int byte_in_row,
@@ -118,13 +120,16 @@ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
   const int head = bid % s.H;
   const int b = bid / s.H;
 
-  const bf16* qbase = q + (long)b * s.sqb + head * s.D;
+  // this block owns keys of batch b; under kv_swap those keys are attended
+  // by batch b^1's queries, so q/do/lse/delta come from b^1
+  const int bq = s.kv_swap ? (b ^ 1) : b;
+  const bf16* qbase = q + (long)bq * s.sqb + head * s.D;
   const bf16* kbase = k + (long)b * s.skb + head * s.D;
   const bf16* vbase = v + (long)b * s.svb + head * s.D;
   const int HD = s.H * s.D;
-  const bf16* dobase = dout + (long)b * s.L * HD + head * s.D;
-  const float* lsebase = lse + (long)b * s.L * s.H + head;
-  const float* delbase = delta + (long)b * s.L * s.H + head;
+  const bf16* dobase = dout + (long)bq * s.L * HD + head * s.D;
+  const float* lsebase = lse + (long)bq * s.L * s.H + head;
+  const float* delbase = delta + (long)bq * s.L * s.H + head;
 
   // ---- preload this wave's K and V rows as A-fragments (rows = keys) ----
   // lane: row = k0 + (l&15), k = dc*32 + (l>>4)*8 .. +8.  K prescaled.
@@ -390,9 +395,11 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ q,
   const int head = bid % s.H;
   const int b = bid / s.H;
 
+  // this block owns queries of batch b; under kv_swap they attend b^1's k/v
+  const int bkv = s.kv_swap ? (b ^ 1) : b;
   const bf16* qbase = q + (long)b * s.sqb + head * s.D;
-  const bf16* kbase = k + (long)b * s.skb + head * s.D;
-  const bf16* vbase = v + (long)b * s.svb + head * s.D;
+  const bf16* kbase = k + (long)bkv * s.skb + head * s.D;
+  const bf16* vbase = v + (long)bkv * s.svb + head * s.D;
   const int HD = s.H * s.D;
   const bf16* dobase = dout + (long)b * s.L * HD + head * s.D;
 
@@ -585,7 +592,7 @@ torch::Tensor attn_delta(torch::Tensor dout, torch::Tensor o) {
 std::vector<torch::Tensor> attn_bwd_fused(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, torch::Tensor dout,
                                           torch::Tensor lse,
-                                          torch::Tensor delta) {
+                                          torch::Tensor delta, bool kv_swap) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   auto check_strides = [](const torch::Tensor& t) {
     TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == t.size(3),
@@ -595,6 +602,9 @@ std::vector<torch::Tensor> attn_bwd_fused(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(dout.is_contiguous() && lse.is_contiguous()
               && delta.is_contiguous());
   ABwdShape s = make_shape(q, k, v);
+  s.kv_swap = kv_swap ? 1 : 0;
+  TORCH_CHECK(!kv_swap || (s.B % 2 == 0 && s.L == s.Lk),
+              "kv_swap pairs batches (b, b^1)");
   TORCH_CHECK(s.L % 128 == 0 && s.Lk % 128 == 0,
               "L must be a multiple of 128");
   TORCH_CHECK(s.D == 16 || s.D == 32 || s.D == 64 || s.D == 128,

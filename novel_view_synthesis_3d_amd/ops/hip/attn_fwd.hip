@@ -44,6 +44,8 @@ struct AttnShape {
   long sqb, sqt;        // q batch, token
   long skb, skt;        // k
   long svb, svt;        // v
+  int kv_swap;          // batch b attends to k/v of batch b^1 (the model's
+                        // cross-FRAME attention with both frames batched)
 };
 
 __device__ __forceinline__ int swz_row(int row, int byte_in_row,
@@ -79,9 +81,10 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
   const int b = bid / s.H;
 
   const int HD = s.H * s.D;  // output layout stays contiguous (B,L,H,D)
+  const int bkv = s.kv_swap ? (b ^ 1) : b;
   const bf16* qbase = q + (long)b * s.sqb + head * s.D;
-  const bf16* kbase = k + (long)b * s.skb + head * s.D;
-  const bf16* vbase = v + (long)b * s.svb + head * s.D;
+  const bf16* kbase = k + (long)bkv * s.skb + head * s.D;
+  const bf16* vbase = v + (long)bkv * s.svb + head * s.D;
 
   // ---- load Q into A-fragment registers, pre-scaled ----
   // chunk dc: lane l holds Q[q0 + (l&15)][dc*32 + (l>>4)*8 .. +8]
@@ -272,7 +275,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q,
 }  // namespace
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v) {
+                                    torch::Tensor v, bool kv_swap) {
   // q/k/v: (B, L, H, D) bf16 — may be strided VIEWS (e.g. into a fused
   // (B, L, 3C) QKV projection) as long as d is contiguous and the head
   // stride is D.
@@ -294,6 +297,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   s.sqb = q.stride(0); s.sqt = q.stride(1);
   s.skb = k.stride(0); s.skt = k.stride(1);
   s.svb = v.stride(0); s.svt = v.stride(1);
+  s.kv_swap = kv_swap ? 1 : 0;
+  TORCH_CHECK(!kv_swap || B % 2 == 0, "kv_swap pairs batches (b, b^1)");
 
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, L, H}, q.options().dtype(torch::kFloat));

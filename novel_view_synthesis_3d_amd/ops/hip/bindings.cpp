@@ -26,13 +26,15 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
                           double out_scale);
 std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
                                          bool with_bias);
+std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
+                                        bool with_bias);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v);
+                                    torch::Tensor v, bool kv_swap);
 torch::Tensor attn_delta(torch::Tensor dout, torch::Tensor o);
 std::vector<torch::Tensor> attn_bwd_fused(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, torch::Tensor dout,
                                           torch::Tensor lse,
-                                          torch::Tensor delta);
+                                          torch::Tensor delta, bool kv_swap);
 torch::Tensor im2col3x3(torch::Tensor x, int64_t stride, int64_t nplanes,
                         int64_t m0, int64_t m1,
                         c10::optional<torch::Tensor> out_buf);
@@ -67,10 +69,11 @@ TORCH_LIBRARY(nvs3d, m) {
         "int step) -> ()");
   m.def("conv3x3_fwd(Tensor x, Tensor w, Tensor? bias, Tensor? residual, float out_scale) -> Tensor");
   m.def("conv3x3_wgrad(Tensor x, Tensor dy, bool with_bias) -> Tensor[]");
-  m.def("attn_fwd(Tensor q, Tensor k, Tensor v) -> Tensor[]");
+  m.def("linear_wgrad(Tensor dy, Tensor x, bool with_bias) -> Tensor[]");
+  m.def("attn_fwd(Tensor q, Tensor k, Tensor v, bool kv_swap=False) -> Tensor[]");
   m.def("attn_delta(Tensor dout, Tensor o) -> Tensor");
   m.def("attn_bwd_fused(Tensor q, Tensor k, Tensor v, Tensor dout, "
-        "Tensor lse, Tensor delta) -> Tensor[]");
+        "Tensor lse, Tensor delta, bool kv_swap=False) -> Tensor[]");
   m.def("im2col3x3(Tensor x, int stride, int nplanes, int m0, int m1, Tensor? out_buf) -> Tensor");
   m.def("attn_p_from_lse(Tensor s, Tensor lse, float scale) -> Tensor");
   m.def("attn_ds(Tensor p, Tensor dp, Tensor delta, float scale) -> Tensor");
@@ -88,6 +91,7 @@ TORCH_LIBRARY_IMPL(nvs3d, CUDA, m) {
   m.impl("fused_adam", fused_adam);
   m.impl("conv3x3_fwd", conv3x3_fwd);
   m.impl("conv3x3_wgrad", conv3x3_wgrad);
+  m.impl("linear_wgrad", linear_wgrad);
   m.impl("attn_fwd", attn_fwd);
   m.impl("attn_delta", attn_delta);
   m.impl("attn_bwd_fused", attn_bwd_fused);

@@ -24,7 +24,8 @@ _OPS = torch.ops.nvs3d
 
 # ops with a HIP implementation (consulted by ops/__init__.py dispatch)
 HAS = {"joint_groupnorm", "pose_embedding", "frame_conv3x3", "attention",
-       "nearest_upsample2x", "avgpool_downsample2x", "residual_scale_add"}
+       "nearest_upsample2x", "avgpool_downsample2x", "residual_scale_add",
+       "linear"}
 
 
 def conv_shapes_supported(cin: int, cout: int, stride: int) -> bool:
@@ -382,10 +383,17 @@ def frame_conv3x3(x, weight, bias, stride: int = 1):
 # backward (rocBLAS batched matmuls — the "plain library GEMM" path).
 # ---------------------------------------------------------------------------
 
+def _swap_frame_pairs(t):
+    """(2B', ...) -> pairs (2i, 2i+1) exchanged (cross-frame kv order)."""
+    B2 = t.shape[0]
+    return (t.reshape(B2 // 2, 2, *t.shape[1:]).flip(1)
+            .reshape(t.shape).contiguous())
+
+
 class _Attention(torch.autograd.Function):
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
-    def forward(ctx, q, k, v):
+    def forward(ctx, q, k, v, kv_swap):
         # strided views (fused-QKV slices) are consumed directly; the
         # kernel requires only a contiguous (head, d) tail
         if q.stride(-1) != 1 or q.stride(2) != q.size(3):
@@ -394,8 +402,9 @@ class _Attention(torch.autograd.Function):
             k = k.contiguous()
         if v.stride(-1) != 1 or v.stride(2) != v.size(3):
             v = v.contiguous()
-        out, lse = _OPS.attn_fwd(q, k, v)
+        out, lse = _OPS.attn_fwd(q, k, v, kv_swap)
         ctx.save_for_backward(q, k, v, out, lse)
+        ctx.kv_swap = kv_swap
         return out
 
     @staticmethod
@@ -404,14 +413,20 @@ class _Attention(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         B, L, H, D = q.shape
         Lk = k.shape[1]
+        swap = ctx.kv_swap
         if (D in (16, 32, 64, 128) and L % 128 == 0 and Lk % 128 == 0
                 and os.environ.get("NVS3D_ATTN_BWD", "fused") != "gemm"):
             # fused flash-style backward (attn_bwd.hip): tile-wise recompute
             # of S/P from q/k/lse — no (B,H,L,L) tensor touches HBM
             doc = do.to(q.dtype).contiguous()
             delta = _OPS.attn_delta(doc, o)
-            dq, dk, dv = _OPS.attn_bwd_fused(q, k, v, doc, lse, delta)
-            return dq, dk, dv
+            dq, dk, dv = _OPS.attn_bwd_fused(q, k, v, doc, lse, delta, swap)
+            return dq, dk, dv, None
+        if swap:
+            # GEMM-recompute fallback shapes (d=256 / short L): materialize
+            # the frame swap, then un-swap the kv grads
+            k = _swap_frame_pairs(k)
+            v = _swap_frame_pairs(v)
         scale = 1.0 / (D ** 0.5)
         # (B,H,L,D) strided views; rocBLAS consumes them without copies
         qt = q.permute(0, 2, 1, 3)
@@ -423,17 +438,70 @@ class _Attention(torch.autograd.Function):
         p = _OPS.attn_p_from_lse(s, lse, scale)        # (B,H,L,Lk) bf16
         dv = torch.matmul(p.transpose(-1, -2), dob)
         dp = torch.matmul(dob, vt.transpose(-1, -2)).contiguous()
-        delta = (do.float() * o.float()).sum(-1).permute(0, 2, 1)  # (B,H,L)
-        ds = _OPS.attn_ds(p, dp, delta.contiguous(), scale)
+        delta = _OPS.attn_delta(do.to(q.dtype).contiguous(), o)
+        ds = _OPS.attn_ds(p, dp, delta.permute(0, 2, 1).contiguous(), scale)
         dq = torch.matmul(ds, kt)
         dk = torch.matmul(ds.transpose(-1, -2), qt)
-        return (dq.permute(0, 2, 1, 3).contiguous(),
-                dk.permute(0, 2, 1, 3).contiguous(),
-                dv.permute(0, 2, 1, 3).contiguous())
+        dq = dq.permute(0, 2, 1, 3).contiguous()
+        dk = dk.permute(0, 2, 1, 3).contiguous()
+        dv = dv.permute(0, 2, 1, 3).contiguous()
+        if swap:
+            dk = _swap_frame_pairs(dk)
+            dv = _swap_frame_pairs(dv)
+        return dq, dk, dv, None
 
 
-def attention(q, k, v):
-    return _Attention.apply(q, k, v)
+def attention(q, k, v, kv_swap: bool = False):
+    return _Attention.apply(q, k, v, kv_swap)
+
+
+# ---------------------------------------------------------------------------
+# Linear (K6): hipBLASLt forward; custom split-K MFMA wgrad for the
+# tall-skinny shapes hipBLASLt collapses on (gemm_wgrad.hip).
+# ---------------------------------------------------------------------------
+
+def _linear_wgrad_supported(M: int, N: int, K: int) -> bool:
+    return (M >= 8192 and N % 8 == 0 and K % 8 == 0
+            and os.environ.get("NVS3D_LINEAR_WGRAD", "mfma") != "blas")
+
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x, w, b):
+        y = torch.nn.functional.linear(x, w, b)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        N, K = w.shape
+        dy = dy.to(x.dtype)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.matmul(dy, w)
+        if ctx.needs_input_grad[1] or ctx.has_bias:
+            dyf = dy.reshape(-1, N).contiguous()
+            xf = x.reshape(-1, K).contiguous()
+            M = dyf.shape[0]
+            if _linear_wgrad_supported(M, N, K):
+                outs = _OPS.linear_wgrad(dyf, xf, ctx.has_bias)
+                dw = outs[0].to(x.dtype)
+                db = outs[1].to(dy.dtype) if ctx.has_bias else None
+            else:
+                dw = torch.matmul(dyf.transpose(0, 1), xf)
+                db = (dyf.sum(0, dtype=torch.float32).to(dy.dtype)
+                      if ctx.has_bias else None)
+            if not ctx.needs_input_grad[1]:
+                dw = None
+        return dx, dw, db
+
+
+def linear(x, w, b=None):
+    return _LinearFn.apply(x, w, b)
 
 
 def frame_conv3x3_residual(x, weight, bias, residual, res_scale):
