@@ -72,8 +72,11 @@ constexpr int WT = 512;    // 8 waves
 struct WgShape {
   int IMG, H, W, Cin, Cout;
   int PW;          // column-tile width (min(W,128)), multiple of 16
+  int pwlog;       // log2(PW)
   int nct;         // column tiles per row = W / PW
-  int sd;          // dy image: bytes per 16-ch subtile (PW*32 + pad)
+  int gr;          // rows staged per barrier round (4/2/1 for PW 16/32/64+)
+  int rs;          // x ring slots (power of 2)
+  int sd;          // dy image: bytes per 16-ch subtile (gr*PW*32 + pad)
   int sx;          // x image: bytes per 16-ch subtile ((PW+2)*32 + pad)
   int units;       // IMG*H*nct row-tiles
   int nb_m, nb_n;  // Cout/WBM, Cin/WBN
@@ -119,9 +122,9 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   if (u0 >= u1) return;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [dy buf 0][dy buf 1][x slot 0..3]  (all offsets arithmetic —
+  // layout: [dy buf 0][dy buf 1][x slot 0..rs-1]  (all offsets arithmetic —
   // no runtime-indexed pointer arrays, §5.4 rule 20)
-  const int dy_bytes = 8 * s.sd;        // 8 subtiles (128 co)
+  const int dy_bytes = 8 * s.sd;        // 8 subtiles (128 co), gr rows each
   const int x_bytes = 4 * s.sx;         // 4 subtiles (64 ci)
   char* xbase = smem + 2 * dy_bytes;
 
@@ -144,8 +147,10 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   const long rowstride_x = (long)s.W * s.Cin;
   const long rowstride_dy = (long)s.W * s.Cout;
 
-  auto stage_dy = [&](int img, int r, int ct, int buf) {
-    const long base = ((long)img * s.H + r) * rowstride_dy
+  // stage `nrows` dy rows [rbase, rbase+nrows) of one (img, ct) into buf:
+  // image pixel index = row_in_group * PW + px
+  auto stage_dy = [&](int img, int rbase, int nrows, int ct, int buf) {
+    const long base = ((long)img * s.H + rbase) * rowstride_dy
                       + (long)ct * s.PW * s.Cout + co0;
     char* dst = smem + buf * dy_bytes;
     const int tot = dy_bytes;
@@ -153,10 +158,15 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
       const int csub = o / s.sd;
       const int w = o % s.sd;
       const bf16* src = zbuf;
-      if (w < s.PW * 32) {
-        const int px = w >> 5;
+      if (w < (s.gr << s.pwlog) * 32) {
+        const int pxf = w >> 5;
+        const int rig = pxf >> s.pwlog;
+        const int px = pxf & (s.PW - 1);
         const int oct = (w & 31) >> 4;
-        src = dy + base + (long)px * s.Cout + csub * 16 + oct * 8;
+        if (rig < nrows) {
+          src = dy + base + (long)rig * rowstride_dy
+                + (long)px * s.Cout + csub * 16 + oct * 8;
+        }
       }
       __builtin_amdgcn_global_load_lds(as_global(src),
           as_shared(dst + (o - lane * 16)), 16, 0, 0);
@@ -193,11 +203,12 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   float dbsum = 0.f;
   const bool do_bias = s.with_bias && bn == 0 && wn == 0;
 
-  // ---- compute one staged row-tile ------------------------------------
-  // dy row r in dybuf[buf]; x rows r-1..r+1 in ring slots (rr+1)&3; the
-  // valid mask enables the ky taps whose x row exists.
-  auto compute_row = [&](int buf, int r, int vmask) {
-    const char* abase = smem + buf * dy_bytes + a_off;
+  // ---- compute one staged row ----------------------------------------
+  // dy row r = group row `ri` of dybuf[buf]; x rows r-1..r+1 in ring slots
+  // (rr+1)&(rs-1); the valid mask enables the ky taps whose x row exists.
+  auto compute_row = [&](int buf, int ri, int r, int vmask) {
+    const char* abase = smem + buf * dy_bytes + a_off
+                        + (ri << s.pwlog) * 32;
 #pragma unroll 1
     for (int p0 = 0; p0 < s.PW; p0 += 16) {
       bf16x8 af = tr16x8(abase + p0 * 32);
@@ -208,7 +219,7 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int ky = 0; ky < 3; ++ky) {
         if (!(vmask & (1 << ky))) continue;
-        const int slot = (r + ky) & 3;   // x row r+ky-1 -> slot (r+ky-1+1)&3
+        const int slot = (r + ky) & (s.rs - 1);  // x row r+ky-1
         const char* bb = xbase + slot * x_bytes + b_off + p0 * 32;
 #pragma unroll
         for (int kx = 0; kx < 3; ++kx) {
@@ -222,10 +233,12 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
     }
   };
 
-  // ---- main loop over row-tiles ---------------------------------------
-  // Segment = maximal run of consecutive rows of one (img, ct): prologue
-  // stages x rows r0-1..r0+1 and dy r0, then each iteration prefetches the
-  // next row's dy and x(r+2) while computing row r.
+  // ---- main loop over row groups --------------------------------------
+  // Segment = maximal run of consecutive rows of one (img, ct), processed
+  // in groups of gr rows per barrier round; the next group's dy + x rows
+  // prefetch under the current group's compute (W<=32 rows are short, so
+  // grouping keeps enough MFMA work between the vmcnt(0)+barrier drains).
+  const int rsm = s.rs - 1;
   int u = u0;
   while (u < u1) {
     const int ct = u / (s.IMG * s.H);
@@ -235,25 +248,34 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
     // segment = run of rows of this (img, ct); img/ct boundaries end it
     const int rend = min(s.H, r0 + (u1 - u));
 
-    // prologue staging for row r0
-    if (r0 > 0) stage_x(img, r0 - 1, ct, r0 & 3);
-    stage_x(img, r0, ct, (r0 + 1) & 3);
-    if (r0 + 1 < s.H) stage_x(img, r0 + 1, ct, (r0 + 2) & 3);
-    stage_dy(img, r0, ct, 0);
+    // prologue: first group's dy rows + x rows [r0-1, r0+cnt]
+    const int cnt0 = min(s.gr, rend - r0);
+    for (int rr = r0 - 1; rr <= r0 + cnt0; ++rr) {
+      if (rr >= 0 && rr < s.H) stage_x(img, rr, ct, (rr + 1) & rsm);
+    }
+    stage_dy(img, r0, cnt0, ct, 0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
-    for (int r = r0; r < rend; ++r) {
-      const int buf = (r - r0) & 1;
-      // prefetch next row's inputs while computing this one
-      if (r + 1 < rend) {
-        stage_dy(img, r + 1, ct, buf ^ 1);
-        if (r + 2 < s.H) stage_x(img, r + 2, ct, (r + 3) & 3);
+    int gbuf = 0;
+    for (int gs = r0; gs < rend; gs += s.gr) {
+      const int cnt = min(s.gr, rend - gs);
+      const int nxt = gs + cnt;
+      if (nxt < rend) {
+        const int ncnt = min(s.gr, rend - nxt);
+        stage_dy(img, nxt, ncnt, ct, gbuf ^ 1);
+        for (int rr = nxt + 1; rr <= nxt + ncnt; ++rr) {
+          if (rr < s.H) stage_x(img, rr, ct, (rr + 1) & rsm);
+        }
       }
-      const int vmask = (r > 0 ? 1 : 0) | 2 | (r + 1 < s.H ? 4 : 0);
-      compute_row(buf, r, vmask);
+      for (int ri = 0; ri < cnt; ++ri) {
+        const int r = gs + ri;
+        const int vmask = (r > 0 ? 1 : 0) | 2 | (r + 1 < s.H ? 4 : 0);
+        compute_row(gbuf, ri, r, vmask);
+      }
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
+      gbuf ^= 1;
     }
     u += rend - r0;
   }
@@ -298,8 +320,12 @@ std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
   TORCH_CHECK(s.W % 16 == 0 && (s.W <= 128 || s.W % 128 == 0));
 
   s.PW = std::min(s.W, 128);
+  s.pwlog = (int)std::round(std::log2((double)s.PW));
+  TORCH_CHECK((1 << s.pwlog) == s.PW, "PW must be a power of two");
   s.nct = s.W / s.PW;
-  s.sd = pad128mod256(s.PW * 32);
+  s.gr = s.PW <= 16 ? 4 : (s.PW <= 32 ? 2 : 1);
+  s.rs = s.gr == 1 ? 4 : (s.gr == 2 ? 8 : 16);
+  s.sd = pad128mod256(s.gr * s.PW * 32);
   s.sx = pad128mod256((s.PW + 2) * 32);
   s.units = s.IMG * s.H * s.nct;
   s.nb_m = s.Cout / WBM;
@@ -320,7 +346,7 @@ std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
     zbuf = torch::zeros({64}, x.options());
   }
 
-  const size_t lds = 2 * (8 * s.sd) + 4 * (4 * s.sx);
+  const size_t lds = 2 * (8 * s.sd) + (size_t)s.rs * (4 * s.sx);
   TORCH_CHECK(lds <= 160 * 1024, "wgrad LDS overflow: ", lds);
   auto stream = at::hip::getCurrentHIPStream();
   const int grid = s.sk * s.nb_m * s.nb_n;

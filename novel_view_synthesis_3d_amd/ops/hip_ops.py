@@ -480,11 +480,13 @@ class _LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         N, K = w.shape
         dy = dy.to(x.dtype)
+        dyf = dy.reshape(-1, N).contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = torch.matmul(dy, w)
+            # ONE flat GEMM (a leading-dim batched matmul would shatter
+            # into thousands of tiny per-row GEMMs on 4/5-D activations)
+            dx = torch.matmul(dyf, w).view(x.shape)
         if ctx.needs_input_grad[1] or ctx.has_bias:
-            dyf = dy.reshape(-1, N).contiguous()
             xf = x.reshape(-1, K).contiguous()
             M = dyf.shape[0]
             if _linear_wgrad_supported(M, N, K):

@@ -52,9 +52,17 @@ def main():
         # wgrad shape: dW = dy^T a  (K_gemm = M)
         dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
         t_wg = timeit(lambda: torch.matmul(dy.transpose(0, 1), a))
+        t_k = timeit(lambda: torch.ops.nvs3d.linear_wgrad(dy, a, True))
+        dw_k = torch.ops.nvs3d.linear_wgrad(dy, a, True)[0]
+        dw_b = torch.matmul(dy.transpose(0, 1).float(), a.float())
+        rel = ((dw_k - dw_b).abs().max() / (dw_b.abs().max() + 1e-6)).item()
         rec = {"op": "dense_wgrad", "tag": tag + "_wgrad",
-               "mnk": [N, K, M], "ms": round(t_wg, 4),
-               "tflops": round(flops / t_wg / 1e9, 1)}
+               "mnk": [N, K, M], "blas_ms": round(t_wg, 4),
+               "mfma_ms": round(t_k, 4),
+               "blas_tflops": round(flops / t_wg / 1e9, 1),
+               "mfma_tflops": round(flops / t_k / 1e9, 1),
+               "speedup": round(t_wg / t_k, 2),
+               "rel_err": round(rel, 5)}
         print(json.dumps(rec), flush=True)
         del a, w, dy
         torch.cuda.empty_cache()

@@ -13,7 +13,7 @@ import torch
 sys.path.insert(0, ".")
 
 
-def run_variant(name, dropout, attn, env=None, steps=6, batch=4):
+def run_variant(name, dropout, attn, env=None, steps=6, batch=4, seed=0):
     import importlib
 
     env = env or {}
@@ -31,7 +31,8 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4):
         if not attn:
             mcfg.attn_resolutions = ()
         tcfg = TrainConfig()
-        tcfg.use_graph = True
+        tcfg.seed = seed
+        tcfg.use_graph = os.environ.get("NVS3D_GRAPH_OFF") != "1"
         tcfg.amp = "bf16"
         tcfg.data = "synthetic"
         losses = []
@@ -59,6 +60,14 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4):
 def main():
     torch.manual_seed(0)
     import sys
+    if len(sys.argv) > 1 and sys.argv[1] == "long":
+        # reproduce bench.py's setup: seed 1234, 14 steps, b16
+        run_variant("graph-b16-s1234-long", dropout=0.1, attn=True,
+                    batch=16, steps=14, seed=1234)
+        run_variant("eager-b16-s1234", dropout=0.1, attn=True,
+                    batch=16, steps=14, seed=1234,
+                    env={"NVS3D_GRAPH_OFF": "1"})
+        return
     if len(sys.argv) > 1 and sys.argv[1] == "batch":
         run_variant("graph-b8", dropout=0.1, attn=True, batch=8)
         run_variant("graph-b12", dropout=0.1, attn=True, batch=12)
