@@ -13,10 +13,12 @@
 // consumed with ds_read_b64_tr_b16 transpose reads (same recipe as
 // conv3x3_wgrad.hip).
 //
-// Geometry: block tile 512(n) x 256(k), 16 waves as 8(n) x 2(k), wave tile
-// 64 x 128 = 8 accumulator planes of mfma_f32_32x32x16_bf16. Grid =
-// NT x KT x SK (512/256-sized n/k tiles; waves whose strip falls outside
-// N/K idle). Chunks of 32 m-rows double-buffered via glds.
+// Geometry: block tile 256(n) x 256(k), 8 waves as 4(n) x 2(k), wave tile
+// 64 x 128 = 8 accumulator planes of mfma_f32_32x32x16_bf16 (8 waves =
+// 2/SIMD so the 128-reg accumulator + operands fit without spilling; a
+// 16-wave block would cap each wave at 128 VGPRs and spill the
+// accumulator to scratch). Grid = NT x KT x SK (waves whose strip falls
+// outside N/K idle). Chunks of 32 m-rows double-buffered via glds.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -48,10 +50,10 @@ __device__ __forceinline__ bf16x8 tr16x8w(const char* p) {
   return __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
 }
 
-constexpr int LBN = 512;   // n per block tile
+constexpr int LBN = 256;   // n per block tile
 constexpr int LBK = 256;   // k per block tile
 constexpr int LMT = 32;    // m rows per staged chunk
-constexpr int LWT = 1024;  // 16 waves
+constexpr int LWT = 512;   // 8 waves
 
 struct LwShape {
   long M;
@@ -87,7 +89,7 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   if (c0 >= c1) return;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int dy_bytes = (LBN / 16) * s.sd;   // 32 subtiles
+  const int dy_bytes = (LBN / 16) * s.sd;   // 16 subtiles
   const int x_bytes = (LBK / 16) * s.sd;    // 16 subtiles
   // layout: [dy buf0][x buf0][dy buf1][x buf1]
   const int buf_bytes = dy_bytes + x_bytes;
@@ -95,7 +97,7 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wn = wave >> 1;   // 0..7: n strip (64 rows)
+  const int wn = wave >> 1;   // 0..3: n strip (64 rows)
   const int wk = wave & 1;    // 0..1: k strip (128 cols)
   const bool live_n = n0 + wn * 64 < s.N;
   const bool live_k = k0 + wk * 128 < s.K;
@@ -108,7 +110,7 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
     char* dst = smem + buf * buf_bytes;
     const long m0 = chunk * LMT;
     // dy chunk -> [csub][m][16]
-    for (int o = wave * 1024 + lane * 16; o < dy_bytes; o += 16 * 1024) {
+    for (int o = wave * 1024 + lane * 16; o < dy_bytes; o += 8 * 1024) {
       const int csub = o / s.sd;
       const int w = o % s.sd;
       const bf16* src = zbuf;
@@ -122,7 +124,7 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
           as_shared(dst + (o - lane * 16)), 16, 0, 0);
     }
     char* dstx = dst + dy_bytes;
-    for (int o = wave * 1024 + lane * 16; o < x_bytes; o += 16 * 1024) {
+    for (int o = wave * 1024 + lane * 16; o < x_bytes; o += 8 * 1024) {
       const int csub = o / s.sd;
       const int w = o % s.sd;
       const bf16* src = zbuf;
@@ -237,7 +239,7 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   s.with_bias = with_bias ? 1 : 0;
   const long chunks = (s.M + LMT - 1) / LMT;
   s.sk = (int)std::max(1L, std::min(chunks,
-                                    (long)(256 / (s.nt * s.kt) + 1)));
+                                    (long)(384 / (s.nt * s.kt) + 1)));
 
   auto opts = x.options().dtype(torch::kFloat);
   auto dw = torch::zeros({s.N, s.K}, opts);

@@ -27,6 +27,7 @@ def timeit(fn, warmup=10, iters=50):
 
 
 def main():
+    from novel_view_synthesis_3d_amd.ops import hip_ops  # loads torch.ops.nvs3d
     # (tag, M, K, N) — b16 full-config shapes
     shapes = [
         ("film_fwd_L0", 16 * 2 * 128 * 128, 256, 512),

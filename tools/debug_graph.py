@@ -36,13 +36,21 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4, seed=0):
         tcfg.amp = "bf16"
         tcfg.data = "synthetic"
         losses = []
+        nosync = os.environ.get("NVS3D_DEBUG_NOSYNC") == "1"
         tr = Trainer(folder=None, train_batch_size=batch,
                      train_num_steps=steps, img_sidelength=128,
                      model_cfg=mcfg, train_cfg=tcfg)
+        kept = []
         for i in range(steps):
             loss = tr.train_step()
+            if nosync:
+                kept.append(loss)  # queue replays back-to-back (bench.py style)
+            else:
+                torch.cuda.synchronize()
+                losses.append(float(loss.item()))
+        if nosync:
             torch.cuda.synchronize()
-            losses.append(float(loss.item()))
+            losses = [float(l.item()) for l in kept[-3:]]
         print(f"[{name}] losses: "
               + " ".join(f"{v:.4f}" for v in losses), flush=True)
         del tr
@@ -67,6 +75,9 @@ def main():
         run_variant("eager-b16-s1234", dropout=0.1, attn=True,
                     batch=16, steps=14, seed=1234,
                     env={"NVS3D_GRAPH_OFF": "1"})
+        run_variant("graph-b16-NOSYNC", dropout=0.1, attn=True,
+                    batch=16, steps=14, seed=1234,
+                    env={"NVS3D_DEBUG_NOSYNC": "1"})
         return
     if len(sys.argv) > 1 and sys.argv[1] == "batch":
         run_variant("graph-b8", dropout=0.1, attn=True, batch=8)
