@@ -451,6 +451,250 @@ void conv3x3_igemm_256(const bf16* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Pipelined 256x256 variant (v2): counted-vmcnt glds across RAW barriers
+// (cdna_hip_programming.md §5.5 T3+T4). The K-tile is consumed in TWO
+// phases (A-quadrants 0-1, then 2-3); staging is piece-granular — a piece
+// is 64 rows x BK of one operand half (8 KB), one glds per wave per piece
+// via the slot->piece chunk map (slot s = piece s, byte s*8192 + wave*1024).
+// Next tile's B pieces issue during phase 0, A pieces (even-row pieces
+// first) during phase 1, so each wait keeps 2-4 glds in flight instead of
+// draining vmcnt(0) at every barrier:
+//   mid-tile:  vmcnt(4)  [this tile's A-odd pieces landed; next B in flight]
+//   boundary:  vmcnt(2)  [next tile's B + A-even landed; A-odd in flight]
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__global__ __launch_bounds__(T2)
+void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
+                          const bf16* __restrict__ w,   // (Cout, 9*Cin)
+                          const float* __restrict__ bias,
+                          const bf16* __restrict__ zbuf,
+                          const bf16* __restrict__ residual,
+                          float out_scale,
+                          bf16* __restrict__ out,
+                          ConvShape s, int nblocks_m) {
+  int bid = blockIdx.x;
+  const int nwg = nblocks_m * (s.Cout / BN2);
+  {
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int bm = bid % nblocks_m;
+  const int bn = bid / nblocks_m;
+  const int m0 = bm * BM2;
+  const int n0 = bn * BN2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto half_base = [&](int buf, int op, int half) -> char* {
+    return smem + (((buf * 2 + op) * 2 + half) << 14);
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  // chunk map: slot s (= piece), half h -> byte o = s*8192 + wave*1024
+  int arow_h[2][2], arow_w[2][2], arow_ok[2][2], kp_[2];  // [half][slot]
+  long arow_base[2][2];
+  int bco[2][2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+#pragma unroll
+    for (int slot = 0; slot < 2; ++slot) {
+      const int o = slot * 8192 + wave * 1024 + lane * 16;
+      const int row = o >> 7;             // 0..127 within the half
+      kp_[slot] = ((o ^ ((row & 7) << 4)) & 127) >> 4;
+      const int m = m0 + half * 128 + row;
+      arow_w[half][slot] = m % s.W;
+      arow_h[half][slot] = (m / s.W) % s.H;
+      arow_ok[half][slot] = m < s.M;
+      arow_base[half][slot] = ((long)(m / (s.W * s.H)) * s.H) * s.W * s.Cin;
+      bco[half][slot] = n0 + half * 128 + row;
+    }
+  }
+
+  // one wave-glds for piece (op, half, slot) of kstep into buf
+  auto stage_piece = [&](int kstep, int buf, int op, int half, int slot) {
+    const int plane = kstep / s.steps_per_plane;
+    const int ci0 = (kstep % s.steps_per_plane) * BK;
+    const int dyp = plane / 3 - 1;
+    const int dxp = plane % 3 - 1;
+    char* dst = half_base(buf, op, half) + slot * 8192 + wave * 1024;
+    if (op == 0) {
+      const int hh = arow_h[half][slot] + dyp;
+      const int ww = arow_w[half][slot] + dxp;
+      const bool valid = arow_ok[half][slot] & (hh >= 0) & (hh < s.H)
+                         & (ww >= 0) & (ww < s.W);
+      const bf16* srcA = valid
+          ? x + arow_base[half][slot] + ((long)hh * s.W + ww) * s.Cin
+              + ci0 + kp_[slot] * 8
+          : zbuf;
+      __builtin_amdgcn_global_load_lds(as_global(srcA), as_shared(dst),
+                                       16, 0, 0);
+    } else {
+      const bf16* srcB = w + (long)bco[half][slot] * (9 * s.Cin)
+                         + kstep * BK + kp_[slot] * 8;
+      __builtin_amdgcn_global_load_lds(as_global(srcB), as_shared(dst),
+                                       16, 0, 0);
+    }
+  };
+  auto stage_op = [&](int kstep, int buf, int op) {
+    // pieces: slot 0 (even rows) of both halves first, then slot 1
+#pragma unroll
+    for (int slot = 0; slot < 2; ++slot)
+#pragma unroll
+      for (int half = 0; half < 2; ++half)
+        stage_piece(kstep, buf, op, half, slot);
+  };
+
+  float acc[8][4][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  // prologue: tile 0 fully staged
+  stage_op(0, 0, 1);
+  stage_op(0, 0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < s.ksteps; ++t) {
+    const int cur = t & 1;
+    const bool more = t + 1 < s.ksteps;
+    const char* baseA = half_base(cur, 0, wm);
+    const char* baseB = half_base(cur, 1, wn >> 1);
+    // B fragments for the whole K-tile (both phases)
+    bf16x8 bfr[4][2];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int row = (wn & 1) * 64 + j * 16 + (lane & 15);
+        bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
+            baseB + swz(row, kk * 32 + (lane >> 4) * 8));
+      }
+    // ---- phase 0: quadrants 0,1 (A piece even); issue next B ----------
+    if (more) stage_op(t + 1, cur ^ 1, 1);
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      bf16x8 afr[2][2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int row = q * 32 + i * 16 + (lane & 15);
+          afr[i][kk] = *reinterpret_cast<const bf16x8*>(
+              baseA + swz(row, kk * 32 + (lane >> 4) * 8));
+        }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]) =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afr[i][kk], bfr[j][kk],
+                    *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]), 0, 0, 0);
+          }
+    }
+    // mid-tile: this tile's A-odd pieces must be visible; next tile's B
+    // (4 glds) stay in flight
+    if (more) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 1: quadrants 2,3 (A piece odd); issue next A -----------
+    if (more) stage_op(t + 1, cur ^ 1, 0);
+#pragma unroll
+    for (int q = 2; q < 4; ++q) {
+      bf16x8 afr[2][2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int row = q * 32 + i * 16 + (lane & 15);
+          afr[i][kk] = *reinterpret_cast<const bf16x8*>(
+              baseA + swz(row, kk * 32 + (lane >> 4) * 8));
+        }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]) =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afr[i][kk], bfr[j][kk],
+                    *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]), 0, 0, 0);
+          }
+    }
+    // boundary: next tile's B + A-even landed; its A-odd (2 glds) in flight
+    if (more) {
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue identical to v1: stage C through LDS, coalesced bf16 stores
+  bf16* ldsC = reinterpret_cast<bf16*>(smem);
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int col = wn * 64 + j * 16 + (lane & 15);
+    const float bj = bias != nullptr ? bias[n0 + col] : 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int rbase = wm * 128 + i * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        ldsC[(rbase + r) * BN2 + col] = __float2bfloat16(acc[i][j][r] + bj);
+      }
+    }
+  }
+  __syncthreads();
+  {
+#pragma unroll
+    for (int it = 0; it < 16; ++it) {
+      const int p = tid + it * T2;
+      const int row = p >> 5;
+      const int cp = p & 31;
+      const int m = m0 + row;
+      if (m < s.M) {
+        Pack<bf16, 8> v = *reinterpret_cast<Pack<bf16, 8>*>(
+            ldsC + row * BN2 + cp * 8);
+        const long goff = (long)m * s.Cout + n0 + cp * 8;
+        if (residual != nullptr) {
+          Pack<bf16, 8> rv = pload<bf16, 8>(residual + goff);
+#pragma unroll
+          for (int jj = 0; jj < 8; ++jj) {
+            from_f32((to_f32(v.v[jj]) + to_f32(rv.v[jj])) * out_scale,
+                     v.v[jj]);
+          }
+        } else if (out_scale != 1.0f) {
+#pragma unroll
+          for (int jj = 0; jj < 8; ++jj) {
+            from_f32(to_f32(v.v[jj]) * out_scale, v.v[jj]);
+          }
+        }
+        pstore<bf16, 8>(out + goff, v);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
@@ -497,7 +741,12 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
       && ((long)(s.M / 256) * (s.Cout / 256) >= 128);
   if (use256) {
     const int nb_m = s.M / BM2;
-    hipLaunchKernelGGL(conv3x3_igemm_256, dim3(nb_m * (s.Cout / BN2)),
+    // pipelined v2 (counted-vmcnt glds across raw barriers) by default;
+    // NVS3D_CONV256=v1 selects the round-1 one-barrier-per-K-tile variant
+    static const char* env = getenv("NVS3D_CONV256");
+    const bool v1 = env && env[0] == 'v' && env[1] == '1';
+    auto kfn = v1 ? conv3x3_igemm_256 : conv3x3_igemm_256_v2;
+    hipLaunchKernelGGL(kfn, dim3(nb_m * (s.Cout / BN2)),
         dim3(T2), 128 * 1024, stream,
         reinterpret_cast<const bf16*>(x.data_ptr()),
         reinterpret_cast<const bf16*>(w.data_ptr()),

@@ -461,7 +461,9 @@ def attention(q, k, v, kv_swap: bool = False):
 # ---------------------------------------------------------------------------
 
 def _linear_wgrad_supported(M: int, N: int, K: int) -> bool:
-    return (M >= 8192 and N % 8 == 0 and K % 8 == 0
+    # measured crossover vs hipBLASLt: the split-K kernel wins decisively at
+    # M >= ~1e5 (2.5-3.4x), ties at 32k, loses below (gpurun_out/dense_bench4)
+    return (M >= 65536 and N % 8 == 0 and K % 8 == 0
             and os.environ.get("NVS3D_LINEAR_WGRAD", "mfma") != "blas")
 
 

@@ -65,6 +65,28 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4, seed=0):
                 os.environ[k] = v
 
 
+def benchlike():
+    """EXACTLY bench.py's structure: 3 unsynced warmup steps, one sync,
+    8 unsynced timed steps, sync, read the loss once."""
+    from novel_view_synthesis_3d_amd.config import TrainConfig, XUNetConfig
+    from novel_view_synthesis_3d_amd.engine.trainer import Trainer
+    tcfg = TrainConfig()
+    tcfg.data = "synthetic"
+    tcfg.amp = "bf16"
+    tcfg.seed = 1234
+    tcfg.use_graph = True
+    tr = Trainer(None, train_batch_size=16, img_sidelength=128,
+                 train_num_steps=10 ** 9, model_cfg=XUNetConfig.named("full"),
+                 train_cfg=tcfg)
+    for _ in range(3):
+        tr.train_step()
+    torch.cuda.synchronize()
+    for _ in range(8):
+        loss = tr.train_step()
+    torch.cuda.synchronize()
+    print(f"[benchlike] final loss {float(loss.item()):.4f}", flush=True)
+
+
 def main():
     torch.manual_seed(0)
     import sys
@@ -78,6 +100,7 @@ def main():
         run_variant("graph-b16-NOSYNC", dropout=0.1, attn=True,
                     batch=16, steps=14, seed=1234,
                     env={"NVS3D_DEBUG_NOSYNC": "1"})
+        benchlike()
         return
     if len(sys.argv) > 1 and sys.argv[1] == "batch":
         run_variant("graph-b8", dropout=0.1, attn=True, batch=8)
