@@ -464,7 +464,6 @@ void conv3x3_igemm_256(const bf16* __restrict__ x,
 //   boundary:  vmcnt(2)  [next tile's B + A-even landed; A-odd in flight]
 // ---------------------------------------------------------------------------
 
-namespace {
 
 __global__ __launch_bounds__(T2)
 void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
