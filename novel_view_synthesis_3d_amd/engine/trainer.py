@@ -248,9 +248,19 @@ class Trainer:
         # capture on the SAME stream the warmup ran on: the parameters' grad
         # accumulators were created on `s` during warmup, and accumulation
         # escapes the capture (-> NaN on replay) if the capture stream
-        # differs from theirs.
-        with torch.cuda.graph(self._graph, stream=s):
-            self._static_loss = self._graph_body()
+        # differs from theirs. Replays then also run on `s`
+        # (_train_step_graphed): replaying a graph on a stream other than
+        # its capture stream raced with the out-of-graph optimizer work and
+        # corrupted the loss buffer when replays were queued back-to-back.
+        import os as _os
+        if _os.environ.get("NVS3D_GRAPH_CAPTURE", "warmup") == "default":
+            self._graph_stream = torch.cuda.current_stream()
+            with torch.cuda.graph(self._graph):
+                self._static_loss = self._graph_body()
+        else:
+            self._graph_stream = s
+            with torch.cuda.graph(self._graph, stream=s):
+                self._static_loss = self._graph_body()
         self.graph_active = True
         if self.is_main:
             print("[trainer] hipGraph captured for the training step",
@@ -272,9 +282,21 @@ class Trainer:
                 raw = self.next_batch()
             for k, v in raw.items():
                 self._static_raw[k].copy_(v, non_blocking=True)
-        self._graph.replay()
-        self.ddp.reduce_all()  # in-place on the flat grad buffers
-        self.opt.step()
+        import os as _os
+        if _os.environ.get("NVS3D_GRAPH_REPLAY", "capture") == "current":
+            self._graph.replay()
+            self.ddp.reduce_all()
+            self.opt.step()
+        else:
+            # replay AND the out-of-graph tail on the capture stream, so
+            # queued replays never interleave with another stream's work
+            cur = torch.cuda.current_stream()
+            self._graph_stream.wait_stream(cur)
+            with torch.cuda.stream(self._graph_stream):
+                self._graph.replay()
+                self.ddp.reduce_all()  # in-place on the flat grad buffers
+                self.opt.step()
+            cur.wait_stream(self._graph_stream)
         return self._static_loss.detach()
 
     def train(self) -> None:

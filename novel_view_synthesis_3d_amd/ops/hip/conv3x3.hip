@@ -740,11 +740,14 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
       && ((long)(s.M / 256) * (s.Cout / 256) >= 128);
   if (use256) {
     const int nb_m = s.M / BM2;
-    // pipelined v2 (counted-vmcnt glds across raw barriers) by default;
-    // NVS3D_CONV256=v1 selects the round-1 one-barrier-per-K-tile variant
+    // v1 (one vmcnt(0)+barrier per K-tile) is the default: the piece-wise
+    // counted-vmcnt v2 MEASURED 5-13% SLOWER on every model shape
+    // (gpurun_out/kb_v1.json vs kb_v2.json — the guide's "coarse
+    // phase-split without the fine per-phase interleave hurts" case);
+    // NVS3D_CONV256=v2 keeps the negative result reproducible
     static const char* env = getenv("NVS3D_CONV256");
-    const bool v1 = env && env[0] == 'v' && env[1] == '1';
-    auto kfn = v1 ? conv3x3_igemm_256 : conv3x3_igemm_256_v2;
+    const bool v2 = env && env[0] == 'v' && env[1] == '2';
+    auto kfn = v2 ? conv3x3_igemm_256_v2 : conv3x3_igemm_256;
     hipLaunchKernelGGL(kfn, dim3(nb_m * (s.Cout / BN2)),
         dim3(T2), 128 * 1024, stream,
         reinterpret_cast<const bf16*>(x.data_ptr()),

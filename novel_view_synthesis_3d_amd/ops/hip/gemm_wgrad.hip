@@ -238,8 +238,10 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   s.sd = pad128mod256w(LMT * 32);
   s.with_bias = with_bias ? 1 : 0;
   const long chunks = (s.M + LMT - 1) / LMT;
+  // 2 blocks/CU (LDS 74KB) so one block's staging stalls hide under the
+  // other's compute — the chunk loop is glds-latency-bound
   s.sk = (int)std::max(1L, std::min(chunks,
-                                    (long)(384 / (s.nt * s.kt) + 1)));
+                                    (long)(512 / (s.nt * s.kt) + 1)));
 
   auto opts = x.options().dtype(torch::kFloat);
   auto dw = torch::zeros({s.N, s.K}, opts);

@@ -65,9 +65,14 @@ def run_variant(name, dropout, attn, env=None, steps=6, batch=4, seed=0):
                 os.environ[k] = v
 
 
-def benchlike():
+def benchlike(tag="", env=None):
     """EXACTLY bench.py's structure: 3 unsynced warmup steps, one sync,
     8 unsynced timed steps, sync, read the loss once."""
+    import time
+    old = {}
+    for k, v in (env or {}).items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
     from novel_view_synthesis_3d_amd.config import TrainConfig, XUNetConfig
     from novel_view_synthesis_3d_amd.engine.trainer import Trainer
     tcfg = TrainConfig()
@@ -81,10 +86,21 @@ def benchlike():
     for _ in range(3):
         tr.train_step()
     torch.cuda.synchronize()
+    t0 = time.perf_counter()
     for _ in range(8):
         loss = tr.train_step()
     torch.cuda.synchronize()
-    print(f"[benchlike] final loss {float(loss.item()):.4f}", flush=True)
+    dt = (time.perf_counter() - t0) / 8
+    wn = float(tr.model.Conv_0.weight.norm().item())
+    print(f"[benchlike{tag}] final loss {float(loss.item()):.4f} "
+          f"wnorm {wn:.4f} ms/step {dt*1e3:.1f}", flush=True)
+    del tr
+    torch.cuda.empty_cache()
+    for k, v in old.items():
+        if v is None:
+            os.environ.pop(k, None)
+        else:
+            os.environ[k] = v
 
 
 def main():
@@ -100,7 +116,10 @@ def main():
         run_variant("graph-b16-NOSYNC", dropout=0.1, attn=True,
                     batch=16, steps=14, seed=1234,
                     env={"NVS3D_DEBUG_NOSYNC": "1"})
-        benchlike()
+        benchlike("-replaycap")  # new default: replay on capture stream
+        benchlike("-replaycur", env={"NVS3D_GRAPH_REPLAY": "current"})
+        benchlike("-capdefault", env={"NVS3D_GRAPH_CAPTURE": "default",
+                                      "NVS3D_GRAPH_REPLAY": "current"})
         return
     if len(sys.argv) > 1 and sys.argv[1] == "batch":
         run_variant("graph-b8", dropout=0.1, attn=True, batch=8)
