@@ -550,6 +550,15 @@ void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
       for (int half = 0; half < 2; ++half)
         stage_piece(kstep, buf, op, half, slot);
   };
+  // one 2-glds step of the per-phase issue schedule: B piece-0s, B
+  // piece-1s, A piece-0s, A piece-1s (phase = which pair)
+  auto stage_phase = [&](int kstep, int buf, int phase) {
+    const int op = phase < 2 ? 1 : 0;
+    const int slot = phase & 1;
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+      stage_piece(kstep, buf, op, half, slot);
+  };
 
   float acc[8][4][4];
 #pragma unroll
@@ -570,7 +579,7 @@ void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
     const bool more = t + 1 < s.ksteps;
     const char* baseA = half_base(cur, 0, wm);
     const char* baseB = half_base(cur, 1, wn >> 1);
-    // B fragments for the whole K-tile (both phases)
+    // B fragments for the whole K-tile (all 4 quadrant phases)
     bf16x8 bfr[4][2];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
@@ -580,10 +589,14 @@ void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
         bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
             baseB + swz(row, kk * 32 + (lane >> 4) * 8));
       }
-    // ---- phase 0: quadrants 0,1 (A piece even); issue next B ----------
-    if (more) stage_op(t + 1, cur ^ 1, 1);
+    // 4 quadrant phases, template-style: {A-quadrant ds_reads | 2-glds
+    // issue | barrier | MFMA under setprio | barrier}; counted vmcnt only
+    // where a consumed piece could still be in flight:
+    //   before phase 2 of this tile: its A piece-1 (issued at phase 3 of
+    //   the previous tile) -> vmcnt(4) [next tile's B pieces in flight]
+    //   at the boundary: next tile's B + A piece-0 -> vmcnt(2)
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
+    for (int q = 0; q < 4; ++q) {
       bf16x8 afr[2][2];
 #pragma unroll
       for (int i = 0; i < 2; ++i)
@@ -593,6 +606,16 @@ void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
           afr[i][kk] = *reinterpret_cast<const bf16x8*>(
               baseA + swz(row, kk * 32 + (lane >> 4) * 8));
         }
+      if (more) stage_phase(t + 1, cur ^ 1, q);
+      if (q == 2) {
+        if (more) {
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -604,41 +627,9 @@ void conv3x3_igemm_256_v2(const bf16* __restrict__ x,
                     afr[i][kk], bfr[j][kk],
                     *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]), 0, 0, 0);
           }
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
     }
-    // mid-tile: this tile's A-odd pieces must be visible; next tile's B
-    // (4 glds) stay in flight
-    if (more) {
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
-    // ---- phase 1: quadrants 2,3 (A piece odd); issue next A -----------
-    if (more) stage_op(t + 1, cur ^ 1, 0);
-#pragma unroll
-    for (int q = 2; q < 4; ++q) {
-      bf16x8 afr[2][2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
-          const int row = q * 32 + i * 16 + (lane & 15);
-          afr[i][kk] = *reinterpret_cast<const bf16x8*>(
-              baseA + swz(row, kk * 32 + (lane >> 4) * 8));
-        }
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-#pragma unroll
-          for (int kk = 0; kk < 2; ++kk) {
-            *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]) =
-                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    afr[i][kk], bfr[j][kk],
-                    *reinterpret_cast<f32x4*>(acc[q * 2 + i][j]), 0, 0, 0);
-          }
-    }
-    // boundary: next tile's B + A-even landed; its A-odd (2 glds) in flight
     if (more) {
       asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     } else {
