@@ -35,6 +35,9 @@ def main() -> None:
     ap.add_argument("--amp", default="bf16", choices=["bf16", "off"])
     ap.add_argument("--graph", default="off", choices=["on", "off"],
                     help="hipGraph-capture the train step (single-node)")
+    ap.add_argument("--bucket-mb", type=float, default=40.0,
+                    help="DDP gradient bucket size (xGMI sweep hook: "
+                         "try 25/40/80)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -48,6 +51,7 @@ def main() -> None:
     train_cfg.amp = args.amp
     train_cfg.seed = 1234
     train_cfg.use_graph = (args.graph == "on" and use_cuda)
+    train_cfg.bucket_mb = args.bucket_mb
 
     trainer = Trainer(None,
                       train_batch_size=args.batch,
@@ -98,6 +102,7 @@ def main() -> None:
                 "global_batch": global_batch,
                 "per_gpu_batch": args.batch,
                 "parallelism": f"dp{n_gpus}",
+                "bucket_mb": args.bucket_mb,
                 "final_loss": float(loss.item()),
                 "params": trainer.model.num_params(),
                 "graph": getattr(trainer, "graph_active", False),
