@@ -113,7 +113,11 @@ class Trainer:
                 ds, batch_size=train_batch_size, shuffle=(sampler is None),
                 sampler=sampler, drop_last=True, collate_fn=ds.collate_fn,
                 pin_memory=(self.device.type == "cuda"),
-                num_workers=cfg.num_workers))
+                num_workers=cfg.num_workers,
+                # without persistent workers a small dataset (epoch = a few
+                # batches) respawns worker processes every epoch — measured
+                # 27 img/s instead of ~700 on the tiny overfit set
+                persistent_workers=(cfg.num_workers > 0)))
         self._gen = torch.Generator(device=self.device)
         self._gen.manual_seed(cfg.seed * 131071 + self.rank + 1)
 

@@ -77,7 +77,7 @@ def overfit_and_sample():
     from PIL import Image
 
     ds = "/tmp/srn64tiny"
-    gen_dataset(ds, instances=3, views=12, size=64, seed=5)
+    gen_dataset(ds, instances=4, views=32, size=64, seed=5)
     tcfg = TrainConfig()
     tcfg.data = "srn"
     tcfg.num_workers = 4
@@ -85,7 +85,7 @@ def overfit_and_sample():
     tcfg.save_every = 10 ** 6
     tcfg.ckpt_folder = "/tmp/ckpt_small"
     mcfg = XUNetConfig.small()
-    steps = int(os.environ.get("NVS3D_EVIDENCE_STEPS", "2500"))
+    steps = int(os.environ.get("NVS3D_EVIDENCE_STEPS", "1500"))
     tr = Trainer(folder=ds, train_batch_size=32, img_sidelength=64,
                  train_num_steps=steps, model_cfg=mcfg, train_cfg=tcfg)
     t0 = time.time()
@@ -153,7 +153,8 @@ def overfit_and_sample():
 
 def main():
     os.makedirs(OUT, exist_ok=True)
-    throughput_srn()
+    if "--sample-only" not in sys.argv:
+        throughput_srn()
     overfit_and_sample()
 
 
