@@ -28,7 +28,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=64,
+    ap.add_argument("--batch", type=int, default=96,
                     help="per-GPU batch size (weak scaling)")
     ap.add_argument("--model", default="full")
     ap.add_argument("--sidelength", type=int, default=128)
