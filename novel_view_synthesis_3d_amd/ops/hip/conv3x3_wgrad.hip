@@ -6,32 +6,31 @@
 //
 // GEMM view per (ky,kx) tap: D[M=co][N=ci] = sum_K A[co][K] B[K][ci] with
 // K = pixels. Both operands are stored pixel-major ([K][C]), so both MFMA
-// fragments need K(pixel)-contiguous elements per lane: the kernel stages
-// dy and x row-tiles into LDS in [csub(16ch)][pixel][16ch-interleave] form
-// and consumes them with ds_read_b64_tr_b16 hardware transpose reads (the
-// attention-V recipe of cdna_hip_programming.md §5.5 T10).
+// fragments need K(pixel)-contiguous elements per lane: row-tiles are
+// staged ROW-MAJOR per pixel ([pixel][C_tile], padded row stride) and
+// consumed with ds_read_b64_tr_b16 hardware transpose reads — the read
+// takes per-lane granule addresses, so the row-major image serves the
+// transposed fragments directly while glds stays line-coalesced, and the
+// row pad spreads each transpose group's 4 pixel-rows over distinct bank
+// classes.
 //
 // Structure:
 //  * One block = (co-tile 128) x (ci-tile 64) x (a contiguous range of
 //    row-tiles). 8 waves as 4(M) x 2(N), each wave one 32x32 output tile
 //    per tap via mfma_f32_32x32x16_bf16 (K-step = 16 pixels).
-//  * ALL NINE (ky,kx) taps accumulate concurrently (9 f32x16 accumulators,
-//    144 regs -> the unified VGPR/AGPR file). The ky dimension pairs the
-//    dy row r with x rows r-1,r,r+1 held in a 4-slot LDS ring; the kx
-//    dimension is a +-1 pixel shift = a +-32 B LDS address offset against
-//    the x row image, which is staged with a 1-pixel halo per row
-//    ([PW+2] pixels; halo = real neighbor pixels of the column tile, or
-//    zero at image edges). So each staged byte of x feeds 9 taps and each
-//    dy byte 9 MFMAs: the kernel reads x and dy from HBM exactly once per
-//    (co,ci)-tile (no 3x/9x im2col materialization).
-//  * Staging is global_load_lds (16 B) with source-permuted addresses
-//    (rule 21): the LDS image is a pure 8-channel-octet permutation of the
-//    global row, so glds stays lane-linear. Subtile stride is padded to
-//    ~= 128 (mod 256) bytes so the 32-lane tr-read service groups land on
-//    disjoint bank halves (conflict-free).
-//  * Split-K: each block atomically adds its fp32 9x32x32-per-wave tile
-//    into dw_acc (Cout,3,3,Cin); bias grad (sum over dy) rides along on
-//    the A fragments of the ci-tile-0 blocks.
+//  * ALL NINE (ky,kx) taps accumulate concurrently (9 f32x16 accumulators).
+//    The ky dimension pairs the dy row r with x rows r-1,r,r+1 held in an
+//    LDS ring; the kx dimension is a +-1 pixel shift = a +-row-stride LDS
+//    offset against the x row image, staged with a 1-pixel halo per row
+//    (halo = real neighbor pixels of the column tile, or zero at image
+//    edges). Each staged byte of x feeds 9 taps: x and dy stream from HBM
+//    exactly once per (co,ci)-tile.
+//  * Rows are processed in groups of gr (4/2/1 for W<=16/32/64+) per
+//    barrier round, prefetching the next group's dy + x rows under the
+//    current group's MFMAs.
+//  * Split-K: each block atomically adds its fp32 9-tap tile into dw_acc
+//    (Cout,3,3,Cin); bias grad rides along on the A fragments of the
+//    ci-tile-0 blocks.
 //
 // Constraints: Cin % 64 == 0, Cout % 128 == 0, W % 16 == 0 and
 // (W <= 128 or W % 128 == 0). Dispatch falls back to the im2col+GEMM path
@@ -59,15 +58,22 @@ __device__ __forceinline__ as3_vp as_shared(void* p) {
   return (as3_vp)(unsigned int)(uintptr_t)p;
 }
 
-__device__ __forceinline__ bf16x8 tr16x8(const char* p) {
-  bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((as3_bf16x4p)(const_cast<char*>(p)));
-  bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((as3_bf16x4p)(const_cast<char*>(p) + 128));
+__device__ __forceinline__ bf16x8 tr16x8(const char* p0, const char* p1) {
+  bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (as3_bf16x4p)(const_cast<char*>(p0)));
+  bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (as3_bf16x4p)(const_cast<char*>(p1)));
   return __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
 }
 
 constexpr int WBM = 128;   // co per block
 constexpr int WBN = 64;    // ci per block
 constexpr int WT = 512;    // 8 waves
+// per-PIXEL row strides (bytes): C_tile*2 + pad. dy: 256+32 (rows step 8
+// banks; group-1 overlap is a minor 2-way on the single A read); x:
+// 128+32 (rows step 40 banks, distinct mod 4).
+constexpr int SROW_DY = WBM * 2 + 32;   // 288
+constexpr int SROW_X = WBN * 2 + 32;    // 160
 
 struct WgShape {
   int IMG, H, W, Cin, Cout;
@@ -76,26 +82,18 @@ struct WgShape {
   int nct;         // column tiles per row = W / PW
   int gr;          // rows staged per barrier round (4/2/1 for PW 16/32/64+)
   int rs;          // x ring slots (power of 2)
-  int sd;          // dy image: bytes per 16-ch subtile (gr*PW*32 + pad)
-  int sx;          // x image: bytes per 16-ch subtile ((PW+2)*32 + pad)
   int units;       // IMG*H*nct row-tiles
   int nb_m, nb_n;  // Cout/WBM, Cin/WBN
   int sk;          // split-K factor
   int with_bias;
 };
 
-// stride pad so that stride % 256 == 128 (disjoint bank halves for the
-// two 16-lane tr-read groups of each 32-lane service group)
-static int pad128mod256(int base) {
-  return base + ((128 - (base % 256)) + 256) % 256;
-}
-
 __global__ __launch_bounds__(WT)
 void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ dy,
                           const bf16* __restrict__ zbuf,
                           float* __restrict__ dwacc,  // (Cout,3,3,Cin) zeroed
-                          float* __restrict__ dbacc,  // (Cout,) zeroed or null
+                          float* __restrict__ dbacc,  // (Cout,) zeroed | null
                           WgShape s) {
   // block decode: bid = sk * (nb_m*nb_n) + (bm*nb_n + bn), with the
   // bijective XCD remap so blocks sharing a split-K data range (identical
@@ -122,10 +120,9 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   if (u0 >= u1) return;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [dy buf 0][dy buf 1][x slot 0..rs-1]  (all offsets arithmetic —
-  // no runtime-indexed pointer arrays, §5.4 rule 20)
-  const int dy_bytes = 8 * s.sd;        // 8 subtiles (128 co), gr rows each
-  const int x_bytes = 4 * s.sx;         // 4 subtiles (64 ci)
+  // layout: [dy buf 0][dy buf 1][x slot 0..rs-1]
+  const int dy_bytes = (s.gr << s.pwlog) * SROW_DY;
+  const int x_bytes = (s.PW + 2) * SROW_X;
   char* xbase = smem + 2 * dy_bytes;
 
   const int tid = threadIdx.x;
@@ -134,38 +131,34 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   const int wm = wave >> 1;   // 0..3: co quarter (32 rows)
   const int wn = wave & 1;    // 0..1: ci half (32 cols)
 
-  // per-lane tr-read base offsets (within an operand image)
-  const int g = lane >> 4;       // 16-lane transpose group
+  // per-lane tr-read base: granule covers channels cbase + (g&1)*16 +
+  // 4*(m16&3) at pixel-row (g>>1)*8 + (m16>>2)
+  const int g = lane >> 4;
   const int m16 = lane & 15;
-  const int lane_px = ((g >> 1) * 8 + (m16 >> 2)) * 32 + (m16 & 3) * 8;
-  const int a_off = (wm * 2 + (g & 1)) * s.sd + lane_px;
-  const int b_off = (wn * 2 + (g & 1)) * s.sx + lane_px;
+  const int tr_prow = (g >> 1) * 8 + (m16 >> 2);
+  const int tr_chb = ((g & 1) * 16 + 4 * (m16 & 3)) * 2;
+  const int a_off = tr_prow * SROW_DY + wm * 64 + tr_chb;
+  const int b_off = tr_prow * SROW_X + wn * 64 + tr_chb;
 
-  // ---- staging (glds, source-permuted) -------------------------------
-  // dy row-tile -> dybuf[buf]; x row-tile -> x slot. Each wave covers LDS
-  // bytes [c*1024 + lane*16) for chunks c = wave, wave+8, ...
   const long rowstride_x = (long)s.W * s.Cin;
   const long rowstride_dy = (long)s.W * s.Cout;
 
   // stage `nrows` dy rows [rbase, rbase+nrows) of one (img, ct) into buf:
-  // image pixel index = row_in_group * PW + px
+  // row-major [group pixel][128 co] (+pad)
   auto stage_dy = [&](int img, int rbase, int nrows, int ct, int buf) {
     const long base = ((long)img * s.H + rbase) * rowstride_dy
                       + (long)ct * s.PW * s.Cout + co0;
     char* dst = smem + buf * dy_bytes;
-    const int tot = dy_bytes;
-    for (int o = wave * 1024 + lane * 16; o < tot; o += 8 * 1024) {
-      const int csub = o / s.sd;
-      const int w = o % s.sd;
+    for (int o = wave * 1024 + lane * 16; o < dy_bytes; o += 8 * 1024) {
+      const int pxf = o / SROW_DY;
+      const int w = o % SROW_DY;
       const bf16* src = zbuf;
-      if (w < (s.gr << s.pwlog) * 32) {
-        const int pxf = w >> 5;
+      if (w < WBM * 2) {
         const int rig = pxf >> s.pwlog;
         const int px = pxf & (s.PW - 1);
-        const int oct = (w & 31) >> 4;
         if (rig < nrows) {
           src = dy + base + (long)rig * rowstride_dy
-                + (long)px * s.Cout + csub * 16 + oct * 8;
+                + (long)px * s.Cout + (w >> 1);
         }
       }
       __builtin_amdgcn_global_load_lds(as_global(src),
@@ -177,16 +170,14 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
     const long base = ((long)img * s.H + r) * rowstride_x + ci0;
     const int px0 = ct * s.PW - 1;  // image pixel of halo index 0
     char* dst = xbase + slot * x_bytes;
-    const int tot = x_bytes;
-    for (int o = wave * 1024 + lane * 16; o < tot; o += 8 * 1024) {
-      const int csub = o / s.sx;
-      const int w = o % s.sx;
+    for (int o = wave * 1024 + lane * 16; o < x_bytes; o += 8 * 1024) {
+      const int pxi = o / SROW_X;
+      const int w = o % SROW_X;
       const bf16* src = zbuf;
-      if (w < (s.PW + 2) * 32) {
-        const int px = px0 + (w >> 5);
-        const int oct = (w & 31) >> 4;
+      if (w < WBN * 2) {
+        const int px = px0 + pxi;
         if (px >= 0 && px < s.W) {
-          src = x + base + (long)px * s.Cin + csub * 16 + oct * 8;
+          src = x + base + (long)px * s.Cin + (w >> 1);
         }
       }
       __builtin_amdgcn_global_load_lds(as_global(src),
@@ -205,13 +196,14 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
 
   // ---- compute one staged row ----------------------------------------
   // dy row r = group row `ri` of dybuf[buf]; x rows r-1..r+1 in ring slots
-  // (rr+1)&(rs-1); the valid mask enables the ky taps whose x row exists.
+  // (rr+1)&(rs-1); vmask enables the ky taps whose x row exists.
   auto compute_row = [&](int buf, int ri, int r, int vmask) {
     const char* abase = smem + buf * dy_bytes + a_off
-                        + (ri << s.pwlog) * 32;
+                        + (ri << s.pwlog) * SROW_DY;
 #pragma unroll 1
     for (int p0 = 0; p0 < s.PW; p0 += 16) {
-      bf16x8 af = tr16x8(abase + p0 * 32);
+      const char* pa = abase + p0 * SROW_DY;
+      bf16x8 af = tr16x8(pa, pa + 4 * SROW_DY);
       if (do_bias) {
 #pragma unroll
         for (int e = 0; e < 8; ++e) dbsum += (float)af[e];
@@ -220,10 +212,12 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
       for (int ky = 0; ky < 3; ++ky) {
         if (!(vmask & (1 << ky))) continue;
         const int slot = (r + ky) & (s.rs - 1);  // x row r+ky-1
-        const char* bb = xbase + slot * x_bytes + b_off + p0 * 32;
+        // x image pixel index = dy pixel + kx (halo offset folded in)
+        const char* bb = xbase + slot * x_bytes + b_off + p0 * SROW_X;
 #pragma unroll
         for (int kx = 0; kx < 3; ++kx) {
-          bf16x8 bf = tr16x8(bb + kx * 32);
+          const char* pb = bb + kx * SROW_X;
+          bf16x8 bf = tr16x8(pb, pb + 4 * SROW_X);
           *reinterpret_cast<f32x16*>(acc[ky * 3 + kx]) =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                   af, bf, *reinterpret_cast<f32x16*>(acc[ky * 3 + kx]),
@@ -234,10 +228,6 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ x,
   };
 
   // ---- main loop over row groups --------------------------------------
-  // Segment = maximal run of consecutive rows of one (img, ct), processed
-  // in groups of gr rows per barrier round; the next group's dy + x rows
-  // prefetch under the current group's compute (W<=32 rows are short, so
-  // grouping keeps enough MFMA work between the vmcnt(0)+barrier drains).
   const int rsm = s.rs - 1;
   int u = u0;
   while (u < u1) {
@@ -325,8 +315,6 @@ std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
   s.nct = s.W / s.PW;
   s.gr = s.PW <= 16 ? 4 : (s.PW <= 32 ? 2 : 1);
   s.rs = s.gr == 1 ? 4 : (s.gr == 2 ? 8 : 16);
-  s.sd = pad128mod256(s.gr * s.PW * 32);
-  s.sx = pad128mod256((s.PW + 2) * 32);
   s.units = s.IMG * s.H * s.nct;
   s.nb_m = s.Cout / WBM;
   s.nb_n = s.Cin / WBN;
@@ -346,7 +334,8 @@ std::vector<torch::Tensor> conv3x3_wgrad(torch::Tensor x, torch::Tensor dy,
     zbuf = torch::zeros({64}, x.options());
   }
 
-  const size_t lds = 2 * (8 * s.sd) + (size_t)s.rs * (4 * s.sx);
+  const size_t lds = 2 * (size_t)((s.gr * s.PW) * SROW_DY)
+                     + (size_t)s.rs * ((s.PW + 2) * SROW_X);
   TORCH_CHECK(lds <= 160 * 1024, "wgrad LDS overflow: ", lds);
   auto stream = at::hip::getCurrentHIPStream();
   const int grid = s.sk * s.nb_m * s.nb_n;

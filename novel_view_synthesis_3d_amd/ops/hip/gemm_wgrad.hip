@@ -7,18 +7,22 @@
 // hipBLASLt collapses on these shapes (N*K tiny, M ~ 10^5-10^6: measured
 // 65-115 TF, >1 ms for the FiLM/skip-Dense wgrads at full config). Here the
 // whole dW tile lives in one block's accumulators and M is split across
-// blocks (fp32 atomic reduction), so dy and x stream from HBM exactly once
-// per (n,k)-tile. Both operands are stored m-major, so both MFMA fragments
-// need m-contiguous lanes: staged in the [csub(16ch)][m][16] LDS layout and
-// consumed with ds_read_b64_tr_b16 transpose reads (same recipe as
-// conv3x3_wgrad.hip).
+// blocks (fp32 atomic reduction), so dy and x stream from HBM once per
+// (n,k)-tile.
+//
+// Both operands are m-major, so both MFMA fragments need m-contiguous
+// lanes: chunks are staged ROW-MAJOR ([m][C_tile], row stride padded +32 B)
+// and consumed with ds_read_b64_tr_b16 — the transpose read takes per-lane
+// granule addresses, so no subtile reshuffle is needed. Row-major staging
+// keeps glds fully line-coalesced (the first version's [csub][m][16]
+// subtile images scattered global reads into 32 B pieces — 1.5 TB/s
+// effective); the +32 B row pad puts the 4 rows of each 16-lane transpose
+// group in distinct bank classes (conflict-free without an XOR swizzle).
 //
 // Geometry: block tile 256(n) x 256(k), 8 waves as 4(n) x 2(k), wave tile
 // 64 x 128 = 8 accumulator planes of mfma_f32_32x32x16_bf16 (8 waves =
-// 2/SIMD so the 128-reg accumulator + operands fit without spilling; a
-// 16-wave block would cap each wave at 128 VGPRs and spill the
-// accumulator to scratch). Grid = NT x KT x SK (waves whose strip falls
-// outside N/K idle). Chunks of 32 m-rows double-buffered via glds.
+// 2/SIMD so the accumulator fits in VGPRs; 16 waves would spill). Grid =
+// NT x KT x SK; chunks of 32 m-rows double-buffered via glds.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -42,11 +46,11 @@ __device__ __forceinline__ as3_vp as_shared(void* p) {
   return (as3_vp)(unsigned int)(uintptr_t)p;
 }
 
-__device__ __forceinline__ bf16x8 tr16x8w(const char* p) {
+__device__ __forceinline__ bf16x8 tr16x8w(const char* p0, const char* p1) {
   bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (as3_bf16x4p)(const_cast<char*>(p)));
+      (as3_bf16x4p)(const_cast<char*>(p0)));
   bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (as3_bf16x4p)(const_cast<char*>(p) + 128));
+      (as3_bf16x4p)(const_cast<char*>(p1)));
   return __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
 }
 
@@ -54,18 +58,15 @@ constexpr int LBN = 256;   // n per block tile
 constexpr int LBK = 256;   // k per block tile
 constexpr int LMT = 32;    // m rows per staged chunk
 constexpr int LWT = 512;   // 8 waves
+// row stride: C_tile*2 bytes + 64 pad (row step = 16 banks: the 32-lane tr service group lands on 32 distinct banks)
+constexpr int LROW = LBN * 2 + 64;
 
 struct LwShape {
   long M;
   int N, K;
   int nt, kt, sk;
-  int sd;          // dy image: bytes per 16-ch subtile (LMT*32 + pad)
   int with_bias;
 };
-
-static int pad128mod256w(int base) {
-  return base + ((128 - (base % 256)) + 256) % 256;
-}
 
 __global__ __launch_bounds__(LWT)
 void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
@@ -89,10 +90,8 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   if (c0 >= c1) return;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int dy_bytes = (LBN / 16) * s.sd;   // 16 subtiles
-  const int x_bytes = (LBK / 16) * s.sd;    // 16 subtiles
-  // layout: [dy buf0][x buf0][dy buf1][x buf1]
-  const int buf_bytes = dy_bytes + x_bytes;
+  constexpr int img_bytes = LMT * LROW;       // one operand image (17.4 KB)
+  constexpr int buf_bytes = 2 * img_bytes;    // [dy][x]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -102,37 +101,40 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   const bool live_n = n0 + wn * 64 < s.N;
   const bool live_k = k0 + wk * 128 < s.K;
 
+  // per-lane tr-read base: granule of lane l covers channels
+  // cbase + (g&1)*16 + 4*(m16&3) at row (g>>1)*8 + (m16>>2) (+ rr*4 + p0)
   const int g = lane >> 4;
   const int m16 = lane & 15;
-  const int lane_px = ((g >> 1) * 8 + (m16 >> 2)) * 32 + (m16 & 3) * 8;
+  const int tr_row = (g >> 1) * 8 + (m16 >> 2);
+  const int tr_chb = ((g & 1) * 16 + 4 * (m16 & 3)) * 2;
+  const int tr_off = tr_row * LROW + tr_chb;
 
   auto stage = [&](long chunk, int buf) {
     char* dst = smem + buf * buf_bytes;
     const long m0 = chunk * LMT;
-    // dy chunk -> [csub][m][16]
-    for (int o = wave * 1024 + lane * 16; o < dy_bytes; o += 8 * 1024) {
-      const int csub = o / s.sd;
-      const int w = o % s.sd;
+    // dy chunk -> row-major [m][LBN] (+pad); lane-linear glds, sources
+    // walk global lines contiguously within each row
+    for (int o = wave * 1024 + lane * 16; o < img_bytes; o += 8 * 1024) {
+      const int m = o / LROW;
+      const int w = o % LROW;
       const bf16* src = zbuf;
-      if (w < LMT * 32) {
-        const long m = m0 + (w >> 5);
-        const int oct = (w & 31) >> 4;
-        const int n = n0 + csub * 16 + oct * 8;
-        if (m < s.M && n < s.N) src = dy + m * s.N + n;
+      if (w < LBN * 2) {
+        const long mm = m0 + m;
+        const int n = n0 + (w >> 1);
+        if (mm < s.M && n < s.N) src = dy + mm * s.N + n;
       }
       __builtin_amdgcn_global_load_lds(as_global(src),
           as_shared(dst + (o - lane * 16)), 16, 0, 0);
     }
-    char* dstx = dst + dy_bytes;
-    for (int o = wave * 1024 + lane * 16; o < x_bytes; o += 8 * 1024) {
-      const int csub = o / s.sd;
-      const int w = o % s.sd;
+    char* dstx = dst + img_bytes;
+    for (int o = wave * 1024 + lane * 16; o < img_bytes; o += 8 * 1024) {
+      const int m = o / LROW;
+      const int w = o % LROW;
       const bf16* src = zbuf;
-      if (w < LMT * 32) {
-        const long m = m0 + (w >> 5);
-        const int oct = (w & 31) >> 4;
-        const int k = k0 + csub * 16 + oct * 8;
-        if (m < s.M && k < s.K) src = x + m * s.K + k;
+      if (w < LBK * 2) {
+        const long mm = m0 + m;
+        const int k = k0 + (w >> 1);
+        if (mm < s.M && k < s.K) src = x + mm * s.K + k;
       }
       __builtin_amdgcn_global_load_lds(as_global(src),
           as_shared(dstx + (o - lane * 16)), 16, 0, 0);
@@ -154,16 +156,16 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   for (long c = c0; c < c1; ++c) {
     const int buf = (int)((c - c0) & 1);
     if (c + 1 < c1) stage(c + 1, buf ^ 1);
-    const char* dyb = smem + buf * buf_bytes;
-    const char* xb = dyb + dy_bytes;
+    const char* dyb = smem + buf * buf_bytes + tr_off;
+    const char* xb = dyb + img_bytes;
     if (live_n) {
 #pragma unroll
       for (int p0 = 0; p0 < LMT; p0 += 16) {
         bf16x8 af[2];
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
-          const int csub = wn * 4 + i * 2 + (g & 1);
-          af[i] = tr16x8w(dyb + csub * s.sd + lane_px + p0 * 32);
+          const char* pa = dyb + p0 * LROW + (wn * 64 + i * 32) * 2;
+          af[i] = tr16x8w(pa, pa + 4 * LROW);
         }
         if (do_bias) {
 #pragma unroll
@@ -174,8 +176,8 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
         if (live_k) {
 #pragma unroll
           for (int j = 0; j < 4; ++j) {
-            const int csub = wk * 8 + j * 2 + (g & 1);
-            bf16x8 bf = tr16x8w(xb + csub * s.sd + lane_px + p0 * 32);
+            const char* pb = xb + p0 * LROW + (wk * 128 + j * 32) * 2;
+            bf16x8 bf = tr16x8w(pb, pb + 4 * LROW);
 #pragma unroll
             for (int i = 0; i < 2; ++i) {
               *reinterpret_cast<f32x16*>(acc[i * 4 + j]) =
@@ -235,7 +237,6 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   TORCH_CHECK(s.N % 8 == 0 && s.K % 8 == 0, "N/K must be 16B-packable");
   s.nt = (s.N + LBN - 1) / LBN;
   s.kt = (s.K + LBK - 1) / LBK;
-  s.sd = pad128mod256w(LMT * 32);
   s.with_bias = with_bias ? 1 : 0;
   const long chunks = (s.M + LMT - 1) / LMT;
   // 2 blocks/CU (LDS 74KB) so one block's staging stalls hide under the
@@ -252,7 +253,7 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   if (!zbuf.defined() || zbuf.device() != x.device()) {
     zbuf = torch::zeros({64}, x.options());
   }
-  const size_t lds = 2 * ((LBN / 16) + (LBK / 16)) * (size_t)s.sd;
+  const size_t lds = 2 * 2 * (size_t)(LMT * LROW);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(linear_wgrad_kernel,
       dim3(s.sk * s.nt * s.kt), dim3(LWT), lds, stream,
