@@ -56,7 +56,6 @@ __device__ __forceinline__ bf16x8 tr16x8w(const char* p0, const char* p1) {
 
 constexpr int LBN = 256;   // n per block tile
 constexpr int LBK = 256;   // k per block tile
-constexpr int LMT = 32;    // m rows per staged chunk
 constexpr int LWT = 512;   // 8 waves
 // row stride: C_tile*2 bytes + 64 pad (row step = 16 banks: the 32-lane tr service group lands on 32 distinct banks)
 constexpr int LROW = LBN * 2 + 64;
@@ -68,6 +67,10 @@ struct LwShape {
   int with_bias;
 };
 
+// LMT (m rows per staged chunk) is a template knob: 32 keeps 2 blocks/CU
+// (74 KB LDS), 64 halves the per-chunk glds-latency stalls at 1 block/CU
+// (147 KB) — the chunk loop is latency-bound, measured A/B picks it.
+template <int LMT>
 __global__ __launch_bounds__(LWT)
 void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
                          const bf16* __restrict__ x,    // (M, K)
@@ -90,7 +93,7 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   if (c0 >= c1) return;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr int img_bytes = LMT * LROW;       // one operand image (17.4 KB)
+  constexpr int img_bytes = LMT * LROW;       // one operand image
   constexpr int buf_bytes = 2 * img_bytes;    // [dy][x]
 
   const int tid = threadIdx.x;
@@ -238,6 +241,8 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   s.nt = (s.N + LBN - 1) / LBN;
   s.kt = (s.K + LBK - 1) / LBK;
   s.with_bias = with_bias ? 1 : 0;
+  static const char* lwgc = getenv("NVS3D_LWG_CHUNK");
+  const int LMT = (lwgc && lwgc[0] == '3') ? 32 : 64;
   const long chunks = (s.M + LMT - 1) / LMT;
   // 2 blocks/CU (LDS 74KB) so one block's staging stalls hide under the
   // other's compute — the chunk loop is glds-latency-bound
@@ -255,7 +260,8 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   }
   const size_t lds = 2 * 2 * (size_t)(LMT * LROW);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(linear_wgrad_kernel,
+  auto kfn = LMT == 32 ? linear_wgrad_kernel<32> : linear_wgrad_kernel<64>;
+  hipLaunchKernelGGL(kfn,
       dim3(s.sk * s.nt * s.kt), dim3(LWT), lds, stream,
       reinterpret_cast<const bf16*>(dy.data_ptr()),
       reinterpret_cast<const bf16*>(x.data_ptr()),
