@@ -99,16 +99,12 @@ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                          ABwdShape s) {
   constexpr int DC = (D + 31) / 32;
   constexpr int DF = (D + 15) / 16;
-  // LDS: Q [64][D] | DO [64][D] | Q^T [D][64] | DO^T [D][64]
-  //      | per-wave P/ds tile (8)[16][64] | lse/delta tiles (64 f32 each)
+  // LDS: 2 x { Q [64][D] | DO [64][D] | Q^T [D][64] | DO^T [D][64] |
+  // lse/delta (64 f32 each) } double-buffered (the next q-tile stages
+  // under this one's MFMAs) | per-wave P/ds tile (8)[16][64]
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* ldsQ = smem;
-  char* ldsDO = ldsQ + QTB * D * 2;
-  char* ldsQT = ldsDO + QTB * D * 2;
-  char* ldsDOT = ldsQT + D * QTB * 2;
-  char* ldsP = ldsDOT + D * QTB * 2;            // 8 waves x 16 x 64 bf16
-  float* ldsLse = reinterpret_cast<float*>(ldsP + 8 * 16 * QTB * 2);
-  float* ldsDel = ldsLse + QTB;
+  constexpr int QSET = 4 * QTB * D * 2 + 2 * QTB * 4;  // one buffer set
+  char* ldsP = smem + 2 * QSET;                 // 8 waves x 16 x 64 bf16
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -161,41 +157,58 @@ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
 
   char* pw = ldsP + wave * (16 * QTB * 2);
 
-  const int qtiles = s.L / QTB;
-  for (int t = 0; t < qtiles; ++t) {
-    __syncthreads();  // previous tile's reads done
-    // ---- stage Q/DO (row-major swz + transposed scatter) + lse/delta ----
-    {
-      constexpr int PACKS = QTB * D / 8;
+  // stage q-tile t into buffer set `buf` (row-major swz + transposed
+  // scatter + lse/delta)
+  auto stage_qtile = [&](int t, int buf) {
+    char* ldsQ = smem + buf * QSET;
+    char* ldsDO = ldsQ + QTB * D * 2;
+    char* ldsQT = ldsDO + QTB * D * 2;
+    char* ldsDOT = ldsQT + D * QTB * 2;
+    float* ldsLse = reinterpret_cast<float*>(ldsDOT + D * QTB * 2);
+    float* ldsDel = ldsLse + QTB;
+    constexpr int PACKS = QTB * D / 8;
 #pragma unroll
-      for (int it = 0; it < (PACKS + 511) / 512; ++it) {
-        const int p = tid + it * 512;
-        if (p < PACKS) {
-          const int row = p / (D / 8);
-          const int d0 = (p % (D / 8)) * 8;
-          Pack<bf16, 8> pq =
-              pload<bf16, 8>(qbase + (long)(t * QTB + row) * s.sqt + d0);
-          *reinterpret_cast<Pack<bf16, 8>*>(
-              ldsQ + swz_row(row, d0 * 2, D * 2)) = pq;
-          Pack<bf16, 8> pdo =
-              pload<bf16, 8>(dobase + (long)(t * QTB + row) * HD + d0);
-          *reinterpret_cast<Pack<bf16, 8>*>(
-              ldsDO + swz_row(row, d0 * 2, D * 2)) = pdo;
+    for (int it = 0; it < (PACKS + 511) / 512; ++it) {
+      const int p = tid + it * 512;
+      if (p < PACKS) {
+        const int row = p / (D / 8);
+        const int d0 = (p % (D / 8)) * 8;
+        Pack<bf16, 8> pq =
+            pload<bf16, 8>(qbase + (long)(t * QTB + row) * s.sqt + d0);
+        *reinterpret_cast<Pack<bf16, 8>*>(
+            ldsQ + swz_row(row, d0 * 2, D * 2)) = pq;
+        Pack<bf16, 8> pdo =
+            pload<bf16, 8>(dobase + (long)(t * QTB + row) * HD + d0);
+        *reinterpret_cast<Pack<bf16, 8>*>(
+            ldsDO + swz_row(row, d0 * 2, D * 2)) = pdo;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            *reinterpret_cast<bf16*>(
-                ldsQT + swz_row(d0 + j, row * 2, QTB * 2)) = pq.v[j];
-            *reinterpret_cast<bf16*>(
-                ldsDOT + swz_row(d0 + j, row * 2, QTB * 2)) = pdo.v[j];
-          }
+        for (int j = 0; j < 8; ++j) {
+          *reinterpret_cast<bf16*>(
+              ldsQT + swz_row(d0 + j, row * 2, QTB * 2)) = pq.v[j];
+          *reinterpret_cast<bf16*>(
+              ldsDOT + swz_row(d0 + j, row * 2, QTB * 2)) = pdo.v[j];
         }
       }
-      if (tid < QTB) {
-        ldsLse[tid] = lsebase[(long)(t * QTB + tid) * s.H];
-        ldsDel[tid] = delbase[(long)(t * QTB + tid) * s.H];
-      }
     }
-    __syncthreads();
+    if (tid < QTB) {
+      ldsLse[tid] = lsebase[(long)(t * QTB + tid) * s.H];
+      ldsDel[tid] = delbase[(long)(t * QTB + tid) * s.H];
+    }
+  };
+
+  const int qtiles = s.L / QTB;
+  stage_qtile(0, 0);
+  __syncthreads();
+  for (int t = 0; t < qtiles; ++t) {
+    const int buf = t & 1;
+    if (t + 1 < qtiles) stage_qtile(t + 1, buf ^ 1);
+    const char* ldsQ = smem + buf * QSET;
+    const char* ldsDO = ldsQ + QTB * D * 2;
+    const char* ldsQT = ldsDO + QTB * D * 2;
+    const char* ldsDOT = ldsQT + D * QTB * 2;
+    const float* ldsLse = reinterpret_cast<const float*>(
+        ldsDOT + D * QTB * 2);
+    const float* ldsDel = ldsLse + QTB;
 
     // ---- S^T[j][i] (rows = keys, cols = q rows), P^T = exp(S^T - lse_i)
     f32x4 st[4];
@@ -297,12 +310,14 @@ void attn_bwd_dkv_kernel(const bf16* __restrict__ q,
                 dsa, qtf, *reinterpret_cast<f32x4*>(dkacc[f]), 0, 0, 0);
       }
     }
+    // next tile's staging writes (buf^1) visible + this buf's reads done
+    // before t+2 overwrites it
+    __syncthreads();
   }
 
   // ---- write dk/dv (C layout row = key, col = d) through LDS ----
-  __syncthreads();
-  // reuse ldsQ region: per wave [16][D]
-  char* ow = ldsQ + wave * (16 * D * 2);
+  // reuse the smem front: per wave [16][D]
+  char* ow = smem + wave * (16 * D * 2);
 #pragma unroll
   for (int f = 0; f < DF; ++f) {
 #pragma unroll
@@ -618,8 +633,8 @@ std::vector<torch::Tensor> attn_bwd_fused(torch::Tensor q, torch::Tensor k,
 
 #define LAUNCH_BWD(DV)                                                     \
   {                                                                        \
-    const size_t lds_kv = (size_t)4 * QTB * DV * 2 + 8 * 16 * QTB * 2      \
-                          + 2 * QTB * 4;                                   \
+    const size_t lds_kv = 2 * ((size_t)4 * QTB * DV * 2 + 2 * QTB * 4)    \
+                          + 8 * 16 * QTB * 2;                              \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<DV>),                          \
         dim3(s.B * s.H * (s.Lk / KVB)), dim3(512), lds_kv, stream,         \
         reinterpret_cast<const bf16*>(q.data_ptr()),                       \
