@@ -220,6 +220,8 @@ WGRAD_SHAPES = [
     (1, 2, 32, 32, 64, 128),    # W=32
     (1, 2, 64, 64, 128, 256),   # L1-ish
     (2, 2, 128, 16, 64, 128),   # W=128 (full-config L0 width), multi-image
+    (1, 2, 16, 256, 64, 128),   # W=256: multiple column tiles (nct=2) with
+                                # real (non-zero) halo pixels at the seam
 ]
 
 
