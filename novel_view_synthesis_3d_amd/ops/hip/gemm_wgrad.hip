@@ -246,8 +246,10 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   const long chunks = (s.M + LMT - 1) / LMT;
   // 2 blocks/CU (LDS 74KB) so one block's staging stalls hide under the
   // other's compute — the chunk loop is glds-latency-bound
+  static const char* lwgs = getenv("NVS3D_LWG_SK");
+  const int target = lwgs ? atoi(lwgs) : 512;
   s.sk = (int)std::max(1L, std::min(chunks,
-                                    (long)(512 / (s.nt * s.kt) + 1)));
+                                    (long)(target / (s.nt * s.kt) + 1)));
 
   auto opts = x.options().dtype(torch::kFloat);
   auto dw = torch::zeros({s.N, s.K}, opts);
