@@ -152,13 +152,31 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
   float dbsum[2] = {0.f, 0.f};
   const bool do_bias = s.with_bias && k0 == 0 && wk == 0 && live_n;
 
+  // Depth-2 glds prefetch over a 3-buffer ring with COUNTED vmcnt across
+  // RAW barriers (the chunk loop is glds-latency-bound; a vmcnt(0) drain
+  // per chunk left it at half roofline). Per-wave glds per chunk is fixed
+  // by the chunk map (the low-id waves cover one extra 1 KB piece per
+  // operand), so the wait that leaves exactly chunk c+1 in flight is a
+  // per-wave constant: iteration c = {wait until only c+1 outstanding;
+  // raw barrier; issue c+2; compute c}. The barrier guarantees every wave
+  // finished computing c-1, so buffer (c+2)%3 == (c-1)%3 is free.
+  const int gper = 2 * ((img_bytes - wave * 1024 + 8191) / 8192);
   stage(c0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  if (c0 + 1 < c1) stage(c0 + 1, 1);
 
   for (long c = c0; c < c1; ++c) {
-    const int buf = (int)((c - c0) & 1);
-    if (c + 1 < c1) stage(c + 1, buf ^ 1);
+    const int buf = (int)((c - c0) % 3);
+    if (c + 1 < c1) {
+      if (gper == 6) {
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      }
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    if (c + 2 < c1) stage(c + 2, (int)((c + 2 - c0) % 3));
     const char* dyb = smem + buf * buf_bytes + tr_off;
     const char* xb = dyb + img_bytes;
     if (live_n) {
@@ -192,8 +210,6 @@ void linear_wgrad_kernel(const bf16* __restrict__ dy,   // (M, N)
         }
       }
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
   }
 
   // epilogue: fp32 atomics. D layout: col = lane&31, row = (e&3) + 8*(e>>2)
@@ -241,8 +257,7 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   s.nt = (s.N + LBN - 1) / LBN;
   s.kt = (s.K + LBK - 1) / LBK;
   s.with_bias = with_bias ? 1 : 0;
-  static const char* lwgc = getenv("NVS3D_LWG_CHUNK");
-  const int LMT = (lwgc && lwgc[0] == '3') ? 32 : 64;
+  const int LMT = 32;  // 3-buffer prefetch ring (110 KB LDS)
   const long chunks = (s.M + LMT - 1) / LMT;
   // 2 blocks/CU (LDS 74KB) so one block's staging stalls hide under the
   // other's compute — the chunk loop is glds-latency-bound
@@ -260,9 +275,9 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor dy, torch::Tensor x,
   if (!zbuf.defined() || zbuf.device() != x.device()) {
     zbuf = torch::zeros({64}, x.options());
   }
-  const size_t lds = 2 * 2 * (size_t)(LMT * LROW);
+  const size_t lds = 3 * 2 * (size_t)(LMT * LROW);
   auto stream = at::hip::getCurrentHIPStream();
-  auto kfn = LMT == 32 ? linear_wgrad_kernel<32> : linear_wgrad_kernel<64>;
+  auto kfn = linear_wgrad_kernel<32>;
   hipLaunchKernelGGL(kfn,
       dim3(s.sk * s.nt * s.kt), dim3(LWT), lds, stream,
       reinterpret_cast<const bf16*>(dy.data_ptr()),
