@@ -11,13 +11,13 @@
 // (n,k)-tile.
 //
 // Both operands are m-major, so both MFMA fragments need m-contiguous
-// lanes: chunks are staged ROW-MAJOR ([m][C_tile], row stride padded +32 B)
+// lanes: chunks are staged ROW-MAJOR ([m][C_tile], row stride padded +64 B)
 // and consumed with ds_read_b64_tr_b16 — the transpose read takes per-lane
 // granule addresses, so no subtile reshuffle is needed. Row-major staging
 // keeps glds fully line-coalesced (the first version's [csub][m][16]
 // subtile images scattered global reads into 32 B pieces — 1.5 TB/s
-// effective); the +32 B row pad puts the 4 rows of each 16-lane transpose
-// group in distinct bank classes (conflict-free without an XOR swizzle).
+// effective); the +64 B row pad puts the rows of each 32-lane transpose
+// service group in distinct bank classes (conflict-free without an XOR swizzle).
 //
 // Geometry: block tile 256(n) x 256(k), 8 waves as 4(n) x 2(k), wave tile
 // 64 x 128 = 8 accumulator planes of mfma_f32_32x32x16_bf16 (8 waves =
