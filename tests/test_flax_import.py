@@ -162,3 +162,25 @@ def test_flax_msgpack_bf16_leaves(tmp_path):
         np.testing.assert_allclose(to_f(p.detach().cpu()).numpy(),
                                    want[path_key].numpy(), rtol=1e-6,
                                    err_msg=path_key)
+
+
+def test_flax_msgpack_strict_mismatch_raises(tmp_path):
+    import msgpack
+    import pytest
+
+    from novel_view_synthesis_3d_amd.engine.checkpoint import (
+        import_flax_msgpack,
+    )
+
+    def ext_pack(x):
+        payload = msgpack.packb((x.shape, x.dtype.name, x.tobytes("C")),
+                                use_bin_type=True)
+        return msgpack.ExtType(1, payload)
+
+    model = _tiny_model()
+    f = tmp_path / "bad_msgpack"
+    f.write_bytes(msgpack.packb(
+        {"not": {"a": {"param": np.zeros(3, dtype=np.float32)}}},
+        default=ext_pack, strict_types=True))
+    with pytest.raises(KeyError):
+        import_flax_msgpack(model, str(f))
